@@ -1,0 +1,187 @@
+#!/usr/bin/env python
+"""Flagship benchmark: BERT-base classification fine-tuning step throughput.
+
+Measures BASELINE.json's metric — samples/sec (whole-job aggregate) + epoch
+wall-clock equivalent — for BERT-base cls seq128 bs32/GPU on 1..8 MI355X,
+synthetic data, random-init weights.
+
+    python bench.py --gpus N --steps K --warmup W
+    # N>1 is launched by the driver as:
+    # python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+    #     --master-addr 127.0.0.1 --master-port P bench.py --gpus N ...
+
+Timing: W untimed warmup steps, then exactly K steps bracketed by
+barrier + torch.cuda.synchronize on both sides; MAX elapsed over ranks;
+rank 0 prints ONE JSON line.
+
+Baseline for ``vs_baseline``: the reference's fastest published number —
+HF Trainer fp16, 0.49 min/epoch on 2 GPUs = 312.9 samples/s (BASELINE.md,
+reference README.md:23), linearly scaled per GPU: baseline(N) = 156.46 * N.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+import torch.distributed as dist
+
+# reference epoch = 9200 train samples; fastest published row:
+# HF Trainer fp16 0.49 min on 2 GPUs
+_REF_EPOCH_SAMPLES = 9200
+_BASELINE_SPS_PER_GPU = _REF_EPOCH_SAMPLES / (0.49 * 60.0) / 2.0  # 156.46
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=30)
+    p.add_argument("--warmup", type=int, default=10)
+    p.add_argument("--model", default="bert-base")
+    p.add_argument("--batch-size", type=int, default=32)
+    p.add_argument("--seq-len", type=int, default=128)
+    p.add_argument("--dtype", default="bf16", choices=["bf16", "fp16", "fp32"])
+    p.add_argument("--bucket-cap-mb", type=float, default=50.0)
+    p.add_argument("--zero", action="store_true", help="ZeRO mode instead of DDP")
+    p.add_argument("--no-ddp-overlap", action="store_true")
+    return p.parse_args()
+
+
+def main():
+    ns = parse_args()
+    world = int(os.environ.get("WORLD_SIZE", 1))
+    rank = int(os.environ.get("RANK", 0))
+    local_rank = int(os.environ.get("LOCAL_RANK", 0))
+    use_cuda = torch.cuda.is_available()
+
+    from pdnlp_amd.utils import set_seed
+    set_seed(123)
+
+    if world > 1:
+        from pdnlp_amd.parallel.bootstrap import init_distributed
+        init_distributed()
+    device = torch.device(f"cuda:{local_rank}" if use_cuda else "cpu")
+    if use_cuda:
+        torch.cuda.set_device(device)
+
+    from pdnlp_amd.models import build_model
+    from pdnlp_amd.ops.adamw import build_optimizer
+    from pdnlp_amd.parallel.ddp import DistributedDataParallel
+    from pdnlp_amd.parallel.zero import ZeroRedundancyOptimizer
+
+    model = build_model(ns.model, model_path=None)
+    dtype = {"bf16": torch.bfloat16, "fp16": torch.float16,
+             "fp32": torch.float32}[ns.dtype]
+    if dtype != torch.float32:
+        model = model.to(dtype)
+    model = model.to(device)
+    model.train()
+
+    scaler = None
+    if ns.dtype == "fp16":
+        from pdnlp_amd.amp import GradScaler
+        scaler = GradScaler()
+
+    if ns.zero and world > 1:
+        optimizer = ZeroRedundancyOptimizer(model, lr=3e-5,
+                                            bucket_mb=ns.bucket_cap_mb)
+        wrapped = model
+    else:
+        optimizer = build_optimizer(model, lr=3e-5)
+        wrapped = model
+        if world > 1:
+            wrapped = DistributedDataParallel(
+                model, bucket_cap_mb=ns.bucket_cap_mb,
+                overlap_comm=not ns.no_ddp_overlap)
+
+    # synthetic batch of the reference shape, resident on device
+    g = torch.Generator().manual_seed(1234 + rank)
+    vocab = model.config.vocab_size if hasattr(model, "config") else 21128
+    ids = torch.randint(106, vocab, (ns.batch_size, ns.seq_len), generator=g)
+    ids[:, 0] = 101
+    ids = ids.to(device)
+    mask = torch.ones_like(ids)
+    type_ids = torch.zeros_like(ids)
+    labels = torch.randint(0, 6, (ns.batch_size,), generator=g).to(device)
+
+    def step():
+        out = wrapped(input_ids=ids, attention_mask=mask,
+                      token_type_ids=type_ids, labels=labels)
+        loss = out.loss
+        if scaler is not None:
+            loss = scaler.scale(loss)
+        loss.backward()
+        if isinstance(wrapped, DistributedDataParallel):
+            wrapped.finalize_backward()
+        if scaler is not None:
+            scaler.step(optimizer)
+            scaler.update()
+        else:
+            optimizer.step()
+        if isinstance(wrapped, DistributedDataParallel):
+            wrapped.zero_grad_buffers()
+        else:
+            optimizer.zero_grad(set_to_none=False)
+        return out.loss
+
+    for _ in range(ns.warmup):
+        step()
+
+    if world > 1:
+        dist.barrier()
+    if use_cuda:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(ns.steps):
+        step()
+    if world > 1:
+        dist.barrier()
+    if use_cuda:
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+
+    if world > 1:
+        t = torch.tensor([elapsed], device=device if use_cuda else "cpu",
+                         dtype=torch.float64)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    n_gpus = world if world > 1 else 1
+    total_samples = ns.batch_size * n_gpus * ns.steps
+    sps = total_samples / elapsed
+    ms_per_step = elapsed / ns.steps * 1000.0
+    baseline = _BASELINE_SPS_PER_GPU * n_gpus
+    if rank == 0:
+        result = {
+            "metric": "samples_per_sec",
+            "value": round(sps, 2),
+            "unit": "samples/s",
+            "n_gpus": n_gpus,
+            "steps": ns.steps,
+            "warmup": ns.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": round(sps / baseline, 3),
+            "dtype": ns.dtype,
+            "data": "synthetic",
+            "config": {
+                "model": ns.model,
+                "global_batch": ns.batch_size * n_gpus,
+                "seq_len": ns.seq_len,
+                "parallelism": ("zero%d" % n_gpus) if ns.zero else f"dp{n_gpus}",
+                "epoch_equiv_min": round(_REF_EPOCH_SAMPLES / sps / 60.0, 4),
+                "baseline_source": "reference README.md:23 HF-Trainer fp16 "
+                                   "0.49 min/epoch on 2 GPUs, scaled per GPU",
+            },
+        }
+        print(json.dumps(result))
+    if world > 1:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,130 @@
+"""Shared CLI pipeline behind the reference-named entrypoint scripts
+(layer L7 of SURVEY.md §1).
+
+Every entrypoint (single-gpu-cls.py, multi-gpu-distributed-cls.py, …) is a
+thin wrapper over ``run_classification`` with a strategy mode — the reference
+instead duplicates this pipeline per script.
+
+Data: uses ``data/train.json`` (same format as the reference) when present;
+otherwise falls back to a synthetic dataset of the same shape so the pipeline
+runs in this no-network environment.
+"""
+
+from __future__ import annotations
+
+import os
+
+import torch
+from torch.utils.data import DataLoader
+
+from .config import Args
+from .data import (ClsDataset, Collate, DistributedSampler, LABELS,
+                   SyntheticClsDataset, build_tokenizer, load_data,
+                   train_dev_split)
+from .engine.trainer import build_training
+from .parallel.bootstrap import cleanup, init_distributed, spawn
+from .utils import set_seed
+from .utils.checkpoint import load_checkpoint
+from .utils.logging import rank0_print
+
+
+_VOCAB = {"tiny": 512, "roberta-base": 50265}
+
+
+def build_dataloaders(args: Args, world_size: int, rank: int):
+    vocab_size = _VOCAB.get(getattr(args, "model", "bert-base"), 21128)
+    if os.path.isfile(args.data_path):
+        data = load_data(args.data_path, limit=args.data_limit)
+        train_data, dev_data = train_dev_split(data, args.ratio, args.seed)
+        tok = build_tokenizer(args.model_path, vocab_size)
+        collate = Collate(tok, args.max_seq_len)
+        train_ds, dev_ds = ClsDataset(train_data), ClsDataset(dev_data)
+    else:
+        rank0_print(f"[data] {args.data_path} not found — synthetic dataset "
+                    f"of the reference shape (seq {args.max_seq_len})")
+        n = args.data_limit
+        n_train = int(n * args.ratio)
+        train_ds = SyntheticClsDataset(n_train, args.max_seq_len, vocab_size,
+                                       seed=args.seed)
+        dev_ds = SyntheticClsDataset(n - n_train, args.max_seq_len, vocab_size,
+                                     seed=args.seed + 1)
+        collate = Collate(None, args.max_seq_len)
+
+    train_sampler = None
+    if world_size > 1:
+        train_sampler = DistributedSampler(train_ds, num_replicas=world_size,
+                                           rank=rank, shuffle=True,
+                                           seed=args.seed)
+    train_loader = DataLoader(
+        train_ds, batch_size=args.train_batch_size,
+        sampler=train_sampler, shuffle=(train_sampler is None),
+        collate_fn=collate, num_workers=args.num_workers,
+        pin_memory=torch.cuda.is_available(), drop_last=False)
+    dev_sampler = None
+    if world_size > 1:
+        dev_sampler = DistributedSampler(dev_ds, num_replicas=world_size,
+                                         rank=rank, shuffle=False)
+    dev_loader = DataLoader(
+        dev_ds, batch_size=args.dev_batch_size, sampler=dev_sampler,
+        shuffle=False, collate_fn=collate, num_workers=args.num_workers,
+        pin_memory=torch.cuda.is_available())
+    return train_loader, dev_loader, train_sampler
+
+
+def run_classification(args: Args, do_test: bool = True):
+    set_seed(args.seed, args.deterministic)
+    args = Args.from_env(args)
+    if args.world_size > 1:
+        args.local_rank = init_distributed()
+        import torch.distributed as dist
+        args.rank = dist.get_rank()
+        args.world_size = dist.get_world_size()
+    elif torch.cuda.is_available():
+        torch.cuda.set_device(args.local_rank)
+
+    rank0_print(f"[pdnlp] strategy={args.strategy} world_size={args.world_size} "
+                f"amp={args.amp}({args.amp_dtype}) device="
+                f"{'cuda' if torch.cuda.is_available() else 'cpu'}")
+
+    train_loader, dev_loader, train_sampler = build_dataloaders(
+        args, args.world_size, args.rank)
+    model, optimizer, scaler, trainer = build_training(args)
+    minutes = trainer.train(train_loader, dev_loader, train_sampler)
+
+    if do_test:
+        # test phase: reload best checkpoint into the bare model, eval on dev
+        # split (reference: multi-gpu-distributed-cls.py:357-365)
+        if os.path.isfile(args.ckpt_path):
+            load_checkpoint(trainer.model, args.ckpt_path,
+                            map_location=trainer.device)
+        trainer.test(dev_loader, label_names=LABELS)
+    if args.world_size > 1:
+        cleanup()
+    return minutes
+
+
+def main(strategy: str = "single", use_spawn: bool = False,
+         amp: bool = False, amp_dtype: str = "bf16", argv=None,
+         extra_defaults: dict = None):
+    args = Args()
+    args.strategy = strategy
+    args.amp = amp
+    args.amp_dtype = amp_dtype
+    for k, v in (extra_defaults or {}).items():
+        setattr(args, k, v)
+    args.apply_cli(argv)
+    if use_spawn:
+        n = args.world_size if args.world_size > 1 else \
+            max(torch.cuda.device_count(), 1)
+        if n > 1:
+            spawn(_spawn_worker, n, args=(args,))
+            return
+        args.world_size = 1
+    run_classification(args)
+
+
+def _spawn_worker(local_rank: int, nprocs: int, args: Args):
+    args.local_rank = local_rank
+    args.rank = local_rank
+    args.world_size = nprocs
+    run_classification(args)

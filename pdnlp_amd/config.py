@@ -1,0 +1,194 @@
+"""Configuration dataclasses.
+
+One dataclass-based config replaces the reference's per-script ``Args`` class
+of plain attributes (reference: multi-gpu-distributed-cls.py:242-257,
+single-gpu-cls.py:193-205) with the same defaults: seq 128, batch 32, lr 3e-5,
+weight-decay 0.01 with no-decay groups, 1 epoch, seed 123, split ratio 0.92.
+"""
+
+from __future__ import annotations
+
+import dataclasses
+import os
+from dataclasses import dataclass, field
+from typing import Optional
+
+
+@dataclass
+class BertConfig:
+    """HF-compatible encoder hyperparameters (BertConfig equivalent)."""
+
+    vocab_size: int = 21128            # chinese-bert-wwm-ext vocab
+    hidden_size: int = 768
+    num_hidden_layers: int = 12
+    num_attention_heads: int = 12
+    intermediate_size: int = 3072
+    hidden_act: str = "gelu"
+    hidden_dropout_prob: float = 0.1
+    attention_probs_dropout_prob: float = 0.1
+    max_position_embeddings: int = 512
+    type_vocab_size: int = 2
+    initializer_range: float = 0.02
+    layer_norm_eps: float = 1e-12
+    num_labels: int = 6                # 6 emotion classes (reference data/train.json)
+    pad_token_id: int = 0
+    model_type: str = "bert"           # "bert" | "roberta"
+
+    @property
+    def head_dim(self) -> int:
+        return self.hidden_size // self.num_attention_heads
+
+    # ---- presets matching BASELINE.json configs ----
+    @staticmethod
+    def bert_base_chinese(num_labels: int = 6) -> "BertConfig":
+        return BertConfig(num_labels=num_labels)
+
+    @staticmethod
+    def bert_large(num_labels: int = 6, vocab_size: int = 21128) -> "BertConfig":
+        return BertConfig(
+            vocab_size=vocab_size, hidden_size=1024, num_hidden_layers=24,
+            num_attention_heads=16, intermediate_size=4096, num_labels=num_labels,
+        )
+
+    @staticmethod
+    def roberta_base(num_labels: int = 6) -> "BertConfig":
+        return BertConfig(
+            vocab_size=50265, type_vocab_size=1, pad_token_id=1,
+            max_position_embeddings=514, num_labels=num_labels,
+            model_type="roberta",
+        )
+
+    @staticmethod
+    def tiny(num_labels: int = 6) -> "BertConfig":
+        """Small config for CPU tests."""
+        return BertConfig(
+            vocab_size=512, hidden_size=64, num_hidden_layers=2,
+            num_attention_heads=4, intermediate_size=128,
+            max_position_embeddings=64, num_labels=num_labels,
+            hidden_dropout_prob=0.0, attention_probs_dropout_prob=0.0,
+        )
+
+    def to_dict(self):
+        return dataclasses.asdict(self)
+
+    @staticmethod
+    def from_dict(d) -> "BertConfig":
+        known = {f.name for f in dataclasses.fields(BertConfig)}
+        return BertConfig(**{k: v for k, v in d.items() if k in known})
+
+
+@dataclass
+class Args:
+    """Training-harness config with the reference's exact defaults
+    (reference: multi-gpu-distributed-cls.py:242-257)."""
+
+    # paths
+    model_path: str = "./model_hub/chinese-bert-wwm-ext"
+    ckpt_path: str = "./output/model.pt"
+    data_path: str = "./data/train.json"
+    output_dir: str = "./output"
+
+    # model / data
+    model: str = "bert-base"           # "bert-base"|"bert-large"|"roberta-base"|"tiny"
+    max_seq_len: int = 128
+    ratio: float = 0.92                # train/dev split
+    data_limit: int = 10000            # reference slices first 10k samples
+    num_workers: int = 2
+
+    # optimization
+    train_batch_size: int = 32         # per GPU
+    dev_batch_size: int = 32
+    epochs: int = 1
+    learning_rate: float = 3e-5
+    weight_decay: float = 0.01
+    adam_beta1: float = 0.9
+    adam_beta2: float = 0.999
+    adam_eps: float = 1e-8
+    max_grad_norm: float = 0.0         # 0 = off (reference does not clip)
+    grad_accum_steps: int = 1
+    optimizer: str = "adamw"           # "adamw" | "sgd" (fabric alt-path)
+    sgd_momentum: float = 0.9
+    lr_scheduler: str = "none"         # "none" | "cosine"
+
+    # precision
+    amp: bool = False
+    amp_dtype: str = "bf16"            # bf16-first on CDNA4; "fp16" keeps GradScaler path
+    init_scale: float = 65536.0
+
+    # eval / logging / checkpoint
+    do_dev: bool = True
+    eval_step: int = 100
+    log_every: int = 1
+    loss_reduce_every: int = 1         # reference reduces the loss scalar every step
+    metrics_jsonl: Optional[str] = None
+
+    # distributed
+    strategy: str = "ddp"              # "single"|"dp"|"ddp"|"zero"|"hooks"
+    backend: Optional[str] = None      # auto: nccl on GPU, gloo on CPU
+    bucket_cap_mb: float = 50.0        # xGMI-tuned default (reference DDP: 25)
+    grad_compression: str = "none"     # "none"|"bf16"|"fp16" (horovod-equiv option)
+    overlap_comm: bool = True
+    zero_stage: int = 1
+    barrier_per_step: bool = False     # debug flag reproducing reference semantics
+    find_unused_parameters: bool = False
+
+    # activation checkpointing (deepspeed-equivalent capability)
+    activation_checkpointing: bool = False
+    checkpoint_cpu_offload: bool = False
+
+    # misc
+    seed: int = 123
+    cudnn_benchmark: bool = False
+    deterministic: bool = False
+
+    # runtime-populated topology (reference mutates these on Args too)
+    local_rank: int = field(default=0)
+    rank: int = field(default=0)
+    world_size: int = field(default=1)
+    device: str = field(default="cpu")
+    total_step: int = field(default=0)
+
+    def asdict(self):
+        return dataclasses.asdict(self)
+
+    @staticmethod
+    def from_env(args: "Args" = None) -> "Args":
+        """Harvest the launcher env contract (MASTER_ADDR/PORT/RANK/WORLD_SIZE/
+        LOCAL_RANK — reference: multi-gpu-distributed-cls.py:275-282)."""
+        a = args or Args()
+        a.rank = int(os.environ.get("RANK", a.rank))
+        a.world_size = int(os.environ.get("WORLD_SIZE", a.world_size))
+        a.local_rank = int(os.environ.get("LOCAL_RANK", a.local_rank))
+        return a
+
+    def apply_cli(self, argv=None) -> "Args":
+        """CLI overrides for every field, plus the legacy ``--local-rank``
+        contract (reference: multi-gpu-distributed-cls.py:377-378)."""
+        import argparse
+
+        p = argparse.ArgumentParser()
+        p.add_argument("--local-rank", "--local_rank", dest="local_rank",
+                       type=int, default=None)
+        p.add_argument("--local_world_size", type=int, default=None)
+        for f in dataclasses.fields(Args):
+            if f.name in ("local_rank",):
+                continue
+            t = f.type if isinstance(f.type, type) else None
+            arg = "--" + f.name.replace("_", "-")
+            if f.type in ("bool", bool):
+                p.add_argument(arg, dest=f.name, type=_str2bool, default=None)
+            elif f.type in ("int", int):
+                p.add_argument(arg, dest=f.name, type=int, default=None)
+            elif f.type in ("float", float):
+                p.add_argument(arg, dest=f.name, type=float, default=None)
+            else:
+                p.add_argument(arg, dest=f.name, type=str, default=None)
+        ns, _ = p.parse_known_args(argv)
+        for k, v in vars(ns).items():
+            if v is not None and hasattr(self, k):
+                setattr(self, k, v)
+        return self
+
+
+def _str2bool(v: str) -> bool:
+    return str(v).lower() in ("1", "true", "yes", "y", "t")

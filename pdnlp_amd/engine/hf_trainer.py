@@ -1,0 +1,129 @@
+"""HF-Trainer-style fully-delegated API (SURVEY.md §2.1 HF Trainer row).
+
+Reference ergonomics (multi-gpu-transformers-cls.py:150-181):
+``TrainingArguments`` (fp16, steps-based eval/save, load-best-at-end) +
+``Trainer(model, args, datasets, collator, compute_metrics).train()``.
+Same surface here, delegating to our engine; collator uses the ``labels``
+key like the reference's HF-Trainer variant
+(multi-gpu-transformers-cls.py:86).
+"""
+
+from __future__ import annotations
+
+import os
+from dataclasses import dataclass, field
+from typing import Callable, Optional
+
+import torch
+from torch.utils.data import DataLoader
+
+from ..config import Args
+from ..data.sampler import DistributedSampler
+from ..parallel.bootstrap import get_rank, init_distributed
+from ..utils.checkpoint import save_checkpoint, load_checkpoint
+from ..utils.logging import rank0_print
+from .trainer import Trainer, build_training
+
+
+@dataclass
+class TrainingArguments:
+    output_dir: str = "./output"
+    per_device_train_batch_size: int = 32
+    per_device_eval_batch_size: int = 32
+    num_train_epochs: int = 1
+    learning_rate: float = 3e-5
+    weight_decay: float = 0.01
+    fp16: bool = False
+    bf16: bool = False
+    evaluation_strategy: str = "steps"   # "no" | "steps"
+    eval_steps: int = 100
+    save_steps: int = 100
+    save_strategy: str = "steps"
+    load_best_model_at_end: bool = True
+    metric_for_best_model: str = "accuracy"
+    logging_steps: int = 10
+    seed: int = 123
+    gradient_accumulation_steps: int = 1
+    dataloader_num_workers: int = 2
+
+
+class HFStyleTrainer:
+    def __init__(self, model, args: TrainingArguments,
+                 train_dataset=None, eval_dataset=None,
+                 data_collator: Optional[Callable] = None,
+                 compute_metrics: Optional[Callable] = None):
+        self.hf_args = args
+        self.model = model
+        self.train_dataset = train_dataset
+        self.eval_dataset = eval_dataset
+        self.data_collator = data_collator
+        self.compute_metrics = compute_metrics
+
+        eargs = Args()
+        eargs.train_batch_size = args.per_device_train_batch_size
+        eargs.dev_batch_size = args.per_device_eval_batch_size
+        eargs.epochs = int(args.num_train_epochs)
+        eargs.learning_rate = args.learning_rate
+        eargs.weight_decay = args.weight_decay
+        eargs.amp = args.fp16 or args.bf16
+        eargs.amp_dtype = "fp16" if args.fp16 else "bf16"
+        eargs.eval_step = args.eval_steps
+        eargs.do_dev = args.evaluation_strategy != "no"
+        eargs.log_every = args.logging_steps
+        eargs.grad_accum_steps = args.gradient_accumulation_steps
+        eargs.seed = args.seed
+        eargs.output_dir = args.output_dir
+        eargs.ckpt_path = os.path.join(args.output_dir, "best_model.pt")
+        eargs.local_rank = init_distributed()
+        eargs.strategy = "ddp"
+        self.args = eargs
+        (self.wrapped, self.optimizer, self.scaler,
+         self.engine) = build_training(eargs, model=model, label_key="labels")
+
+    def _loader(self, dataset, batch_size, shuffle):
+        sampler = None
+        import torch.distributed as dist
+        if dist.is_initialized():
+            sampler = DistributedSampler(dataset, shuffle=shuffle)
+        return DataLoader(dataset, batch_size=batch_size,
+                          sampler=sampler, shuffle=(sampler is None and shuffle),
+                          collate_fn=self.data_collator,
+                          num_workers=self.hf_args.dataloader_num_workers), sampler
+
+    def train(self):
+        loader, sampler = self._loader(
+            self.train_dataset, self.hf_args.per_device_train_batch_size, True)
+        eval_loader = None
+        if self.eval_dataset is not None:
+            eval_loader, _ = self._loader(
+                self.eval_dataset, self.hf_args.per_device_eval_batch_size, False)
+        minutes = self.engine.train(loader, eval_loader, sampler)
+        if (self.hf_args.load_best_model_at_end
+                and os.path.isfile(self.args.ckpt_path)):
+            load_checkpoint(self.model, self.args.ckpt_path,
+                            map_location=self.engine.device)
+        return {"train_runtime_min": minutes}
+
+    def evaluate(self):
+        eval_loader, _ = self._loader(
+            self.eval_dataset, self.hf_args.per_device_eval_batch_size, False)
+        loss, acc = self.engine.dev(eval_loader)
+        metrics = {"eval_loss": loss, "eval_accuracy": acc}
+        rank0_print(metrics)
+        return metrics
+
+    def predict(self, dataset):
+        loader, _ = self._loader(
+            dataset, self.hf_args.per_device_eval_batch_size, False)
+        self.engine.model.eval()
+        outs = []
+        with torch.no_grad():
+            for batch in loader:
+                _, logits, _ = self.engine.on_step(batch)
+                outs.append(logits.float().cpu())
+        return torch.cat(outs, dim=0)
+
+    def save_model(self, path: Optional[str] = None):
+        save_checkpoint(self.model,
+                        path or os.path.join(self.hf_args.output_dir, "model.pt"),
+                        rank=get_rank())

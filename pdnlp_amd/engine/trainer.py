@@ -1,0 +1,323 @@
+"""The Trainer loop (layer L4) — one engine with strategy modes, replacing the
+reference's per-script Trainer copies (canonical:
+multi-gpu-distributed-cls.py:113-239).
+
+Behavioral parity with the reference:
+- epoch loop → ``sampler.set_epoch`` → step loop → forward/loss → backward →
+  step → reduced global-mean loss → rank-0 ``【train】 epoch：e/E step：s/S
+  loss：x`` print → periodic dev() with best-acc checkpoint save.
+- dev/test: per-batch loss all-reduce + logits/labels all-gather, CPU argmax
+  accuracy, final sklearn classification_report in test().
+
+Deliberate deviations (SURVEY.md §2.2 notes):
+- no per-step ``dist.barrier()`` in the hot loop (correctness does not need
+  it; ``barrier_per_step`` flag restores reference semantics for debugging);
+- the loss scalar all-reduce can be batched (``loss_reduce_every``);
+- checkpoints are saved UNWRAPPED (no ``module.`` prefix);
+- the AMP path calls ``zero_grad`` every step (the reference's AMP script
+  forgot it — SURVEY.md §3.2 — we fix, not replicate).
+"""
+
+from __future__ import annotations
+
+import time
+from typing import Optional
+
+import numpy as np
+import torch
+import torch.distributed as dist
+
+from ..parallel.ddp import DistributedDataParallel
+from ..parallel.zero import ZeroRedundancyOptimizer
+from ..utils.checkpoint import save_checkpoint
+from ..utils.logging import rank0_print
+from ..utils.metrics import MetricsWriter, StepTimer, TraceRange
+
+
+class Trainer:
+    def __init__(self, args, model, optimizer, device,
+                 scaler=None, lr_scheduler=None, label_key: str = "label"):
+        self.args = args
+        self.model = model
+        self.optimizer = optimizer
+        self.device = torch.device(device)
+        self.scaler = scaler
+        self.lr_scheduler = lr_scheduler
+        self.label_key = label_key
+        self.rank = dist.get_rank() if dist.is_initialized() else 0
+        self.world = dist.get_world_size() if dist.is_initialized() else 1
+        self.metrics = MetricsWriter(getattr(args, "metrics_jsonl", None),
+                                     rank=self.rank)
+        self.best_acc = 0.0
+        self.global_step = 0
+
+    # ------------------------------------------------------------------
+    def on_step(self, batch):
+        """H2D copy + forward (reference: multi-gpu-distributed-cls.py:126-137)."""
+        input_ids = batch["input_ids"].to(self.device, non_blocking=True)
+        attention_mask = batch["attention_mask"].to(self.device, non_blocking=True)
+        token_type_ids = batch["token_type_ids"].to(self.device, non_blocking=True)
+        labels = batch[self.label_key].to(self.device, non_blocking=True)
+        out = self.model(input_ids=input_ids, attention_mask=attention_mask,
+                         token_type_ids=token_type_ids, labels=labels)
+        return out.loss, out.logits, labels
+
+    def loss_reduce(self, loss: torch.Tensor) -> torch.Tensor:
+        """Global mean loss (reference: multi-gpu-distributed-cls.py:139-143)."""
+        if self.world > 1:
+            loss = loss.detach().clone()
+            dist.all_reduce(loss, op=dist.ReduceOp.SUM)
+            loss = loss / self.world
+        return loss
+
+    def output_reduce(self, logits: torch.Tensor, labels: torch.Tensor):
+        """All-gather eval logits+labels (reference:
+        multi-gpu-distributed-cls.py:145-155)."""
+        if self.world == 1:
+            return logits, labels
+        lg = [torch.zeros_like(logits) for _ in range(self.world)]
+        lb = [torch.zeros_like(labels) for _ in range(self.world)]
+        dist.all_gather(lg, logits.contiguous())
+        dist.all_gather(lb, labels.contiguous())
+        return torch.cat(lg, dim=0), torch.cat(lb, dim=0)
+
+    # ------------------------------------------------------------------
+    def _backward_and_step(self, loss, accum_boundary: bool):
+        args = self.args
+        scale_accum = 1.0 / max(args.grad_accum_steps, 1)
+        is_ddp = isinstance(self.model, DistributedDataParallel)
+        is_zero = isinstance(self.optimizer, ZeroRedundancyOptimizer)
+
+        def _bw(l):
+            l = l * scale_accum
+            if self.scaler is not None:
+                l = self.scaler.scale(l)
+            with TraceRange("backward"):
+                l.backward()
+
+        if is_ddp and not accum_boundary:
+            with self.model.no_sync():
+                _bw(loss)
+            return False
+        _bw(loss)
+        if is_ddp:
+            self.model.finalize_backward()
+        if args.max_grad_norm and args.max_grad_norm > 0:
+            if self.scaler is not None:
+                self.scaler.unscale_(self.optimizer)
+            torch.nn.utils.clip_grad_norm_(
+                (p for p in self.model.parameters() if p.grad is not None),
+                args.max_grad_norm)
+        with TraceRange("optimizer"):
+            if self.scaler is not None and not is_zero:
+                self.scaler.step(self.optimizer)
+                self.scaler.update()
+            elif self.scaler is not None and is_zero:
+                self.scaler.unscale_(self.optimizer)
+                if not self.scaler._found_inf:
+                    self.optimizer.step()
+                self.scaler.update()
+            else:
+                self.optimizer.step()
+        if self.lr_scheduler is not None:
+            self.lr_scheduler.step()
+        self._zero_grad()
+        return True
+
+    def _zero_grad(self):
+        if isinstance(self.model, DistributedDataParallel):
+            self.model.zero_grad_buffers()
+        else:
+            self.optimizer.zero_grad(set_to_none=False)
+
+    # ------------------------------------------------------------------
+    def train(self, train_loader, dev_loader=None, train_sampler=None):
+        args = self.args
+        total_step = len(train_loader) * args.epochs
+        args.total_step = total_step
+        self.model.train()
+        timer = StepTimer(self.device)
+        t_start = time.time()
+        micro = 0
+        for epoch in range(1, args.epochs + 1):
+            if train_sampler is not None and hasattr(train_sampler, "set_epoch"):
+                train_sampler.set_epoch(epoch)
+            for step, batch in enumerate(train_loader, start=1):
+                micro += 1
+                accum_boundary = (micro % max(args.grad_accum_steps, 1) == 0)
+                with TraceRange("forward"):
+                    loss, logits, labels = self.on_step(batch)
+                if args.barrier_per_step and self.world > 1:
+                    dist.barrier()
+                stepped = self._backward_and_step(loss, accum_boundary)
+                self.global_step += 1
+                timer.step(labels.shape[0] * self.world)
+                if self.global_step % args.loss_reduce_every == 0:
+                    printed_loss = self.loss_reduce(loss).item()
+                else:
+                    printed_loss = loss.item()
+                if self.global_step % args.log_every == 0:
+                    rank0_print(
+                        f"【train】 epoch：{epoch}/{args.epochs} "
+                        f"step：{self.global_step}/{total_step} "
+                        f"loss：{printed_loss:.6f}")
+                    self.metrics.write(
+                        phase="train", epoch=epoch, step=self.global_step,
+                        loss=printed_loss,
+                        lr=self.optimizer.param_groups[0]["lr"]
+                        if hasattr(self.optimizer, "param_groups") else args.learning_rate,
+                        samples_per_sec=timer.samples_per_sec(),
+                        stepped=stepped)
+                if (dev_loader is not None and args.do_dev
+                        and self.global_step % args.eval_step == 0):
+                    dev_loss, acc = self.dev(dev_loader)
+                    rank0_print(f"【dev】 loss：{dev_loss:.6f} accuracy：{acc:.4f}")
+                    self.metrics.write(phase="dev", step=self.global_step,
+                                       loss=dev_loss, accuracy=acc)
+                    if acc > self.best_acc:
+                        self.best_acc = acc
+                        save_checkpoint(self.model, args.ckpt_path, rank=self.rank)
+                        rank0_print(f"【best】 accuracy：{acc:.4f} → saved "
+                                    f"{args.ckpt_path}")
+                    self.model.train()
+        if not (args.do_dev and dev_loader is not None):
+            save_checkpoint(self.model, args.ckpt_path, rank=self.rank)
+        mins = (time.time() - t_start) / 60.0
+        rank0_print(f"耗时：{mins:.4f}分钟 (wall-clock minutes)")
+        self.metrics.write(phase="train_end", minutes=mins,
+                           samples_per_sec=timer.samples_per_sec(sync=True))
+        return mins
+
+    # ------------------------------------------------------------------
+    @torch.no_grad()
+    def dev(self, loader):
+        """Eval loop (reference: multi-gpu-distributed-cls.py:199-220)."""
+        self.model.eval()
+        total_loss = 0.0
+        preds, trues = [], []
+        for batch in loader:
+            loss, logits, labels = self.on_step(batch)
+            total_loss += self.loss_reduce(loss).item()
+            g_logits, g_labels = self.output_reduce(logits.float(), labels)
+            preds.append(g_logits.argmax(-1).cpu().numpy())
+            trues.append(g_labels.cpu().numpy())
+        preds = np.concatenate(preds)
+        trues = np.concatenate(trues)
+        acc = float((preds == trues).mean()) if len(trues) else 0.0
+        return total_loss, acc
+
+    @torch.no_grad()
+    def test(self, loader, label_names=None):
+        """Test with classification report (reference:
+        multi-gpu-distributed-cls.py:222-239)."""
+        self.model.eval()
+        total_loss = 0.0
+        preds, trues = [], []
+        for batch in loader:
+            loss, logits, labels = self.on_step(batch)
+            total_loss += self.loss_reduce(loss).item()
+            g_logits, g_labels = self.output_reduce(logits.float(), labels)
+            preds.append(g_logits.argmax(-1).cpu().numpy())
+            trues.append(g_labels.cpu().numpy())
+        preds = np.concatenate(preds)
+        trues = np.concatenate(trues)
+        report = classification_report_text(trues, preds, label_names)
+        if self.rank == 0:
+            print(report)
+        acc = float((preds == trues).mean()) if len(trues) else 0.0
+        return total_loss, acc, report
+
+
+def classification_report_text(trues, preds, label_names=None) -> str:
+    try:
+        from sklearn.metrics import classification_report
+        labels = list(range(len(label_names))) if label_names else None
+        return classification_report(trues, preds, labels=labels,
+                                     target_names=label_names,
+                                     zero_division=0)
+    except ImportError:
+        # minimal fallback: per-class P/R/F1
+        import collections
+        lines = ["label\tprec\trecall\tf1\tsupport"]
+        classes = sorted(set(list(trues) + list(preds)))
+        for c in classes:
+            tp = int(((preds == c) & (trues == c)).sum())
+            fp = int(((preds == c) & (trues != c)).sum())
+            fn = int(((preds != c) & (trues == c)).sum())
+            p = tp / (tp + fp) if tp + fp else 0.0
+            r = tp / (tp + fn) if tp + fn else 0.0
+            f1 = 2 * p * r / (p + r) if p + r else 0.0
+            name = label_names[c] if label_names else str(c)
+            lines.append(f"{name}\t{p:.2f}\t{r:.2f}\t{f1:.2f}\t{tp + fn}")
+        return "\n".join(lines)
+
+
+def build_training(args, model=None, label_key: str = "label"):
+    """Assemble (model, optimizer, scaler, trainer) for a strategy mode.
+
+    Strategy seam L3 of SURVEY.md §1: "single" | "dp" | "ddp" | "zero" |
+    "hooks" — one engine, per-mode wiring.
+    """
+    from ..amp import GradScaler, cast_model_to
+    from ..models import build_model
+    from ..ops.adamw import build_optimizer
+    from ..parallel import (DataParallel, DistributedOptimizer,
+                            broadcast_optimizer_state, broadcast_parameters)
+
+    device = torch.device(f"cuda:{args.local_rank}"
+                          if torch.cuda.is_available() else "cpu")
+    args.device = str(device)
+    if model is None:
+        model = build_model(getattr(args, "model", "bert-base"),
+                            model_path=args.model_path)
+    if args.amp:
+        model = cast_model_to(model, args.amp_dtype)
+    model = model.to(device)
+    if getattr(args, "activation_checkpointing", False):
+        model.gradient_checkpointing_enable(
+            cpu_offload=getattr(args, "checkpoint_cpu_offload", False))
+
+    scaler = None
+    if args.amp and args.amp_dtype == "fp16":
+        scaler = GradScaler(init_scale=args.init_scale)
+
+    if args.strategy == "zero":
+        optimizer = ZeroRedundancyOptimizer(
+            model, lr=args.learning_rate,
+            betas=(args.adam_beta1, args.adam_beta2), eps=args.adam_eps,
+            weight_decay=args.weight_decay, bucket_mb=args.bucket_cap_mb)
+        wrapped = model
+    else:
+        optimizer = build_optimizer(
+            model, lr=args.learning_rate, weight_decay=args.weight_decay,
+            betas=(args.adam_beta1, args.adam_beta2), eps=args.adam_eps,
+            optimizer=args.optimizer, sgd_momentum=args.sgd_momentum)
+        if args.strategy == "ddp" and dist.is_initialized():
+            wrapped = DistributedDataParallel(
+                model, bucket_cap_mb=args.bucket_cap_mb,
+                grad_compression=args.grad_compression,
+                overlap_comm=args.overlap_comm)
+        elif args.strategy == "hooks" and dist.is_initialized():
+            broadcast_parameters(model)
+            comp = {"none": None, "bf16": torch.bfloat16,
+                    "fp16": torch.float16}[args.grad_compression]
+            optimizer = DistributedOptimizer(optimizer, compression=comp,
+                                             fusion_mb=args.bucket_cap_mb)
+            broadcast_optimizer_state(optimizer.optimizer)
+            wrapped = model
+        elif args.strategy == "dp":
+            wrapped = DataParallel(model)
+        else:
+            wrapped = model
+
+    lr_scheduler = None
+    if args.lr_scheduler == "cosine":
+        base = optimizer.optimizer if isinstance(optimizer, DistributedOptimizer) \
+            else optimizer
+        if isinstance(base, torch.optim.Optimizer):
+            lr_scheduler = torch.optim.lr_scheduler.CosineAnnealingLR(
+                base, T_max=max(args.total_step, 1) or 1000)
+
+    trainer = Trainer(args, wrapped, optimizer, device, scaler=scaler,
+                      lr_scheduler=lr_scheduler, label_key=label_key)
+    return wrapped, optimizer, scaler, trainer

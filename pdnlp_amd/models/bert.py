@@ -1,0 +1,354 @@
+"""BERT / RoBERTa sequence classifiers with HF checkpoint layout, computed
+through pdnlp_amd.ops (hand-written CDNA4 kernels on GPU).
+
+State-dict keys match HF ``BertForSequenceClassification`` exactly
+(``bert.embeddings.word_embeddings.weight`` …
+``bert.encoder.layer.N.attention.self.query.weight`` … ``classifier.weight``)
+so checkpoints interchange with the reference workflow
+(BASELINE.json: "the same HF BertForSequenceClassification checkpoint
+layout"; reference model build at multi-gpu-distributed-cls.py:336-338).
+
+Parameters live in standard ``nn.Linear``/``nn.Embedding``/``nn.LayerNorm``
+containers (for the key names); forward bypasses their ``forward`` and calls
+the fused ops directly: embedding+LN (K1), fused-QKV GEMM (K2), attention
+(K3-K5), GEMM+bias+GELU (K7), GEMM + bias+dropout+residual+LN epilogues
+(K6/K8), pooler/classifier (K9), CE (K10).
+"""
+
+from __future__ import annotations
+
+import math
+from dataclasses import dataclass
+from typing import Optional
+
+import torch
+import torch.nn as nn
+
+from .. import ops
+from ..config import BertConfig
+
+
+@dataclass
+class SequenceClassifierOutput:
+    loss: Optional[torch.Tensor]
+    logits: torch.Tensor
+
+    def __getitem__(self, i):  # reference indexes output[1] for logits
+        return (self.loss, self.logits)[i]
+
+    def __iter__(self):
+        return iter((self.loss, self.logits))
+
+
+class BertEmbeddings(nn.Module):
+    def __init__(self, cfg: BertConfig):
+        super().__init__()
+        self.cfg = cfg
+        self.word_embeddings = nn.Embedding(cfg.vocab_size, cfg.hidden_size,
+                                            padding_idx=cfg.pad_token_id)
+        self.position_embeddings = nn.Embedding(cfg.max_position_embeddings,
+                                                cfg.hidden_size)
+        self.token_type_embeddings = nn.Embedding(cfg.type_vocab_size,
+                                                  cfg.hidden_size)
+        self.LayerNorm = nn.LayerNorm(cfg.hidden_size, eps=cfg.layer_norm_eps)
+        self.register_buffer(
+            "position_ids",
+            torch.arange(cfg.max_position_embeddings).unsqueeze(0),
+            persistent=False)
+
+    def forward(self, input_ids, token_type_ids, training: bool):
+        B, S = input_ids.shape
+        if self.cfg.model_type == "roberta":
+            # HF roberta: position ids from non-pad positions, offset by pad id
+            mask = (input_ids != self.cfg.pad_token_id).long()
+            position_ids = mask.cumsum(-1) * mask + self.cfg.pad_token_id
+        else:
+            position_ids = self.position_ids[:, :S].expand(B, S)
+        if token_type_ids is None:
+            token_type_ids = torch.zeros_like(input_ids)
+        h = ops.embedding_layernorm(
+            input_ids, token_type_ids, position_ids,
+            self.word_embeddings.weight, self.position_embeddings.weight,
+            self.token_type_embeddings.weight,
+            self.LayerNorm.weight, self.LayerNorm.bias, self.cfg.layer_norm_eps)
+        return ops.dropout(h, self.cfg.hidden_dropout_prob, training)
+
+
+class BertSelfAttention(nn.Module):
+    def __init__(self, cfg: BertConfig):
+        super().__init__()
+        self.query = nn.Linear(cfg.hidden_size, cfg.hidden_size)
+        self.key = nn.Linear(cfg.hidden_size, cfg.hidden_size)
+        self.value = nn.Linear(cfg.hidden_size, cfg.hidden_size)
+
+
+class BertSelfOutput(nn.Module):
+    def __init__(self, cfg: BertConfig):
+        super().__init__()
+        self.dense = nn.Linear(cfg.hidden_size, cfg.hidden_size)
+        self.LayerNorm = nn.LayerNorm(cfg.hidden_size, eps=cfg.layer_norm_eps)
+
+
+class BertAttention(nn.Module):
+    def __init__(self, cfg: BertConfig):
+        super().__init__()
+        self.self = BertSelfAttention(cfg)
+        self.output = BertSelfOutput(cfg)
+
+
+class BertIntermediate(nn.Module):
+    def __init__(self, cfg: BertConfig):
+        super().__init__()
+        self.dense = nn.Linear(cfg.hidden_size, cfg.intermediate_size)
+
+
+class BertOutput(nn.Module):
+    def __init__(self, cfg: BertConfig):
+        super().__init__()
+        self.dense = nn.Linear(cfg.intermediate_size, cfg.hidden_size)
+        self.LayerNorm = nn.LayerNorm(cfg.hidden_size, eps=cfg.layer_norm_eps)
+
+
+class BertLayer(nn.Module):
+    def __init__(self, cfg: BertConfig):
+        super().__init__()
+        self.cfg = cfg
+        self.attention = BertAttention(cfg)
+        self.intermediate = BertIntermediate(cfg)
+        self.output = BertOutput(cfg)
+
+    def forward(self, h: torch.Tensor, attn_mask: Optional[torch.Tensor],
+                training: bool) -> torch.Tensor:
+        cfg = self.cfg
+        B, S, H = h.shape
+        nh, hd = cfg.num_attention_heads, cfg.head_dim
+        a = self.attention.self
+        # fused QKV projection: one [H, 3H] GEMM (K2)
+        wqkv = torch.cat([a.query.weight, a.key.weight, a.value.weight], dim=0)
+        bqkv = torch.cat([a.query.bias, a.key.bias, a.value.bias], dim=0)
+        qkv = ops.linear(h, wqkv, bqkv)
+        q, k, v = qkv.split(H, dim=-1)
+        q = q.view(B, S, nh, hd).transpose(1, 2)
+        k = k.view(B, S, nh, hd).transpose(1, 2)
+        v = v.view(B, S, nh, hd).transpose(1, 2)
+        ctx = ops.attention(q, k, v, attn_mask,
+                            cfg.attention_probs_dropout_prob, training)
+        ctx = ctx.transpose(1, 2).reshape(B, S, H)
+        # attention output projection + fused bias/dropout/residual/LN (K6)
+        ao = self.attention.output
+        proj = ops.linear(ctx, ao.dense.weight, None)
+        h = ops.bias_dropout_residual_layernorm(
+            proj, ao.dense.bias, h, ao.LayerNorm.weight, ao.LayerNorm.bias,
+            cfg.hidden_dropout_prob, training, cfg.layer_norm_eps)
+        # FFN: GEMM + fused GELU (K7), GEMM + fused epilogue (K8)
+        inter = ops.linear(h, self.intermediate.dense.weight,
+                           self.intermediate.dense.bias, act="gelu")
+        down = ops.linear(inter, self.output.dense.weight, None)
+        h = ops.bias_dropout_residual_layernorm(
+            down, self.output.dense.bias, h,
+            self.output.LayerNorm.weight, self.output.LayerNorm.bias,
+            cfg.hidden_dropout_prob, training, cfg.layer_norm_eps)
+        return h
+
+
+class BertEncoder(nn.Module):
+    def __init__(self, cfg: BertConfig):
+        super().__init__()
+        self.layer = nn.ModuleList(BertLayer(cfg)
+                                   for _ in range(cfg.num_hidden_layers))
+
+
+class BertPooler(nn.Module):
+    def __init__(self, cfg: BertConfig):
+        super().__init__()
+        self.dense = nn.Linear(cfg.hidden_size, cfg.hidden_size)
+
+
+class BertModel(nn.Module):
+    def __init__(self, cfg: BertConfig, add_pooler: bool = True):
+        super().__init__()
+        self.cfg = cfg
+        self.embeddings = BertEmbeddings(cfg)
+        self.encoder = BertEncoder(cfg)
+        self.pooler = BertPooler(cfg) if add_pooler else None
+        self._grad_ckpt = False
+        self._ckpt_cpu_offload = False
+
+    def forward(self, input_ids, attention_mask=None, token_type_ids=None):
+        training = self.training
+        if attention_mask is None:
+            attention_mask = torch.ones_like(input_ids)
+        # additive mask in compute dtype: 0 keep / -10000 drop (HF convention)
+        dtype = self.embeddings.word_embeddings.weight.dtype
+        addmask = (1.0 - attention_mask[:, None, None, :].to(dtype)) * -10000.0
+        h = self.embeddings(input_ids, token_type_ids, training)
+        for layer in self.encoder.layer:
+            if self._grad_ckpt and training:
+                h = _checkpoint_layer(layer, h, addmask, training,
+                                      self._ckpt_cpu_offload)
+            else:
+                h = layer(h, addmask, training)
+        pooled = None
+        if self.pooler is not None:
+            pooled = ops.linear(h[:, 0], self.pooler.dense.weight,
+                                self.pooler.dense.bias, act="tanh")
+        return h, pooled
+
+    def gradient_checkpointing_enable(self, cpu_offload: bool = False):
+        self._grad_ckpt = True
+        self._ckpt_cpu_offload = cpu_offload
+
+    def gradient_checkpointing_disable(self):
+        self._grad_ckpt = False
+
+
+def _checkpoint_layer(layer, h, mask, training, cpu_offload):
+    """Activation checkpointing (deepspeed-capability equivalent,
+    reference config: multi-gpu-deepspeed-cls.py:240-244), optional CPU
+    offload of the saved input over PCIe."""
+    import torch.utils.checkpoint as cp
+    if cpu_offload:
+        ctx = torch.autograd.graph.save_on_cpu(pin_memory=True)
+        with ctx:
+            return cp.checkpoint(layer, h, mask, training, use_reentrant=False)
+    return cp.checkpoint(layer, h, mask, training, use_reentrant=False)
+
+
+class BertForSequenceClassification(nn.Module):
+    def __init__(self, cfg: BertConfig):
+        super().__init__()
+        self.cfg = self.config = cfg
+        self.bert = BertModel(cfg, add_pooler=True)
+        self.dropout = nn.Dropout(cfg.hidden_dropout_prob)
+        self.classifier = nn.Linear(cfg.hidden_size, cfg.num_labels)
+        self.apply(_init_weights(cfg))
+
+    def forward(self, input_ids, attention_mask=None, token_type_ids=None,
+                labels=None):
+        _, pooled = self.bert(input_ids, attention_mask, token_type_ids)
+        pooled = ops.dropout(pooled, self.cfg.hidden_dropout_prob, self.training)
+        logits = ops.linear(pooled, self.classifier.weight, self.classifier.bias)
+        loss = ops.cross_entropy(logits, labels) if labels is not None else None
+        return SequenceClassifierOutput(loss=loss, logits=logits)
+
+    def gradient_checkpointing_enable(self, cpu_offload: bool = False):
+        self.bert.gradient_checkpointing_enable(cpu_offload)
+
+    @staticmethod
+    def from_pretrained(path: str, cfg: Optional[BertConfig] = None):
+        return _from_pretrained(BertForSequenceClassification, path, cfg)
+
+
+class RobertaClassificationHead(nn.Module):
+    def __init__(self, cfg: BertConfig):
+        super().__init__()
+        self.dense = nn.Linear(cfg.hidden_size, cfg.hidden_size)
+        self.out_proj = nn.Linear(cfg.hidden_size, cfg.num_labels)
+
+
+class RobertaForSequenceClassification(nn.Module):
+    """RoBERTa-base variant (BASELINE.json config 5): HF ``roberta.*`` keys,
+    no pooler, CLS-token classification head."""
+
+    def __init__(self, cfg: Optional[BertConfig] = None):
+        super().__init__()
+        cfg = cfg or BertConfig.roberta_base()
+        assert cfg.model_type == "roberta"
+        self.cfg = self.config = cfg
+        self.roberta = BertModel(cfg, add_pooler=False)
+        self.classifier = RobertaClassificationHead(cfg)
+        self.apply(_init_weights(cfg))
+
+    def forward(self, input_ids, attention_mask=None, token_type_ids=None,
+                labels=None):
+        h, _ = self.roberta(input_ids, attention_mask, token_type_ids)
+        x = ops.dropout(h[:, 0], self.cfg.hidden_dropout_prob, self.training)
+        x = ops.linear(x, self.classifier.dense.weight,
+                       self.classifier.dense.bias, act="tanh")
+        x = ops.dropout(x, self.cfg.hidden_dropout_prob, self.training)
+        logits = ops.linear(x, self.classifier.out_proj.weight,
+                            self.classifier.out_proj.bias)
+        loss = ops.cross_entropy(logits, labels) if labels is not None else None
+        return SequenceClassifierOutput(loss=loss, logits=logits)
+
+    def gradient_checkpointing_enable(self, cpu_offload: bool = False):
+        self.roberta.gradient_checkpointing_enable(cpu_offload)
+
+    @staticmethod
+    def from_pretrained(path: str, cfg: Optional[BertConfig] = None):
+        return _from_pretrained(RobertaForSequenceClassification, path,
+                                cfg or BertConfig.roberta_base())
+
+
+def _init_weights(cfg: BertConfig):
+    def fn(m):
+        if isinstance(m, nn.Linear):
+            m.weight.data.normal_(0.0, cfg.initializer_range)
+            if m.bias is not None:
+                m.bias.data.zero_()
+        elif isinstance(m, nn.Embedding):
+            m.weight.data.normal_(0.0, cfg.initializer_range)
+            if m.padding_idx is not None:
+                m.weight.data[m.padding_idx].zero_()
+        elif isinstance(m, nn.LayerNorm):
+            m.weight.data.fill_(1.0)
+            m.bias.data.zero_()
+    return fn
+
+
+def _from_pretrained(cls, path: str, cfg: Optional[BertConfig]):
+    """Load from an HF checkpoint dir (config.json + pytorch_model.bin /
+    model.safetensors) or a bare state-dict .pt file."""
+    import json
+    import os
+
+    if os.path.isdir(path):
+        cfg_file = os.path.join(path, "config.json")
+        if cfg is None and os.path.isfile(cfg_file):
+            with open(cfg_file) as f:
+                cfg = BertConfig.from_dict(json.load(f))
+        model = cls(cfg or BertConfig())
+        sd = None
+        st = os.path.join(path, "model.safetensors")
+        pt = os.path.join(path, "pytorch_model.bin")
+        if os.path.isfile(st):
+            from safetensors.torch import load_file
+            sd = load_file(st)
+        elif os.path.isfile(pt):
+            sd = torch.load(pt, map_location="cpu", weights_only=False)
+        if sd is not None:
+            from ..utils.checkpoint import strip_module_prefix
+            missing, unexpected = model.load_state_dict(
+                strip_module_prefix(sd), strict=False)
+            dropped = [k for k in unexpected if "position_ids" not in k]
+            if dropped:
+                raise RuntimeError(f"unexpected checkpoint keys: {dropped[:8]}")
+        return model
+    model = cls(cfg or BertConfig())
+    if os.path.isfile(path):
+        from ..utils.checkpoint import load_checkpoint
+        load_checkpoint(model, path)
+    return model
+
+
+def build_model(name: str = "bert-base", num_labels: int = 6,
+                model_path: Optional[str] = None):
+    """Model factory by preset name (bench + CLI entrypoints)."""
+    import os
+    if name in ("bert-base", "bert-base-chinese"):
+        cfg = BertConfig.bert_base_chinese(num_labels)
+        cls = BertForSequenceClassification
+    elif name == "bert-large":
+        cfg = BertConfig.bert_large(num_labels)
+        cls = BertForSequenceClassification
+    elif name == "roberta-base":
+        cfg = BertConfig.roberta_base(num_labels)
+        cls = RobertaForSequenceClassification
+    elif name == "tiny":
+        cfg = BertConfig.tiny(num_labels)
+        cls = BertForSequenceClassification
+    else:
+        raise ValueError(f"unknown model preset {name}")
+    if model_path and os.path.exists(model_path):
+        return cls.from_pretrained(model_path, cfg)
+    return cls(cfg)

@@ -1,0 +1,85 @@
+"""Op dispatch layer.
+
+Every hot op of the BERT pipeline goes through this package. Two paths:
+
+- **HIP path** (MI355X, gfx950): hand-written CDNA4 kernels in the in-tree
+  extension ``pdnlp_amd/ops/_hip_ext*.so`` (built by ``setup.py build_ext
+  --inplace`` / ``__graft_entry__.build()``). This is the path that runs on
+  a GPU box; if the extension is missing there we raise instead of silently
+  falling back to eager torch.
+- **torch path** (CPU tests, and explicit opt-in via
+  ``PDNLP_ALLOW_TORCH_FALLBACK=1``): plain differentiable torch compositions
+  used as the numerics reference.
+
+The reference has no first-party kernels at all — its hot path is vendor CUDA
+code inside HF BERT; these kernels are the MI355X-native
+equivalents K1-K16.
+"""
+
+from __future__ import annotations
+
+import os
+
+import torch
+
+_EXT = None
+_EXT_ERR = None
+
+
+def _try_load_ext():
+    global _EXT, _EXT_ERR
+    if _EXT is not None or _EXT_ERR is not None:
+        return _EXT
+    try:
+        from . import _hip_ext  # built in-tree; travels with the repo snapshot
+        _EXT = _hip_ext
+    except ImportError:
+        try:
+            import importlib
+            _EXT = importlib.import_module("pdnlp_amd_hip_ext")
+        except ImportError as e:
+            _EXT_ERR = e
+            _EXT = None
+    return _EXT
+
+
+def ext():
+    """The HIP extension module, or raise if on GPU without it."""
+    m = _try_load_ext()
+    if m is None and torch.cuda.is_available() and not allow_fallback():
+        raise RuntimeError(
+            "pdnlp_amd HIP extension is not built but a GPU is visible. "
+            "Build it in-tree with `python setup.py build_ext --inplace` "
+            "(PYTORCH_ROCM_ARCH=gfx950) or set PDNLP_ALLOW_TORCH_FALLBACK=1 "
+            f"to run eager torch ops. Import error: {_EXT_ERR}"
+        )
+    return m
+
+
+def allow_fallback() -> bool:
+    return os.environ.get("PDNLP_ALLOW_TORCH_FALLBACK", "0") == "1"
+
+
+def hip_enabled(t: torch.Tensor = None) -> bool:
+    """True when the HIP kernels should run for this tensor."""
+    if os.environ.get("PDNLP_FORCE_TORCH", "0") == "1":
+        return False
+    if t is not None and t.device.type != "cuda":
+        return False
+    if t is None and not torch.cuda.is_available():
+        return False
+    return ext() is not None
+
+
+from .functional import (  # noqa: F401,E402
+    layernorm,
+    embedding_layernorm,
+    linear,
+    bias_gelu,
+    attention,
+    masked_softmax,
+    cross_entropy,
+    bias_dropout_residual_layernorm,
+    dropout,
+)
+from .adamw import FusedAdamW, multi_tensor_adamw  # noqa: F401,E402

@@ -1,0 +1,153 @@
+"""Fused multi-tensor AdamW (K12) + the reference's optimizer construction.
+
+Reference: ``build_optimizer`` creates AdamW lr=3e-5 with two param groups —
+weight_decay=0.01 for everything except ``bias``/``LayerNorm.weight``
+(single-gpu-cls.py:86-97). Here the update itself is a hand-written
+multi-tensor HIP kernel (one launch per chunk of tensors, not one per tensor),
+with fp32 master weights when params are bf16/fp16, honoring decoupled weight
+decay per group. Torch fallback uses ``torch._foreach_*``.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Iterable, List, Optional
+
+import torch
+
+from . import hip_enabled, ext
+
+CHUNK = 512  # tensors per multi-tensor launch
+
+
+def multi_tensor_adamw(params: List[torch.Tensor], grads: List[torch.Tensor],
+                       exp_avgs: List[torch.Tensor], exp_avg_sqs: List[torch.Tensor],
+                       masters: List[Optional[torch.Tensor]],
+                       lr: float, beta1: float, beta2: float, eps: float,
+                       weight_decay: float, step: int,
+                       grad_scale_inv: float = 1.0) -> None:
+    """Apply one AdamW step to a flat list of tensors (same group)."""
+    if not params:
+        return
+    bc1 = 1.0 - beta1 ** step
+    bc2 = 1.0 - beta2 ** step
+    if hip_enabled(params[0]) and getattr(ext(), "multi_tensor_adamw", None) is not None:
+        use_master = masters[0] is not None
+        for i in range(0, len(params), CHUNK):
+            ext().multi_tensor_adamw(
+                params[i:i + CHUNK], grads[i:i + CHUNK],
+                exp_avgs[i:i + CHUNK], exp_avg_sqs[i:i + CHUNK],
+                masters[i:i + CHUNK] if use_master else [],
+                lr, beta1, beta2, eps, weight_decay, bc1, bc2, grad_scale_inv)
+        return
+    # torch fallback (also the numerics reference)
+    with torch.no_grad():
+        for p, g, m, v, mw in zip(params, grads, exp_avgs, exp_avg_sqs, masters):
+            w = mw if mw is not None else p
+            gf = g.float() * grad_scale_inv
+            if weight_decay != 0.0:
+                w.mul_(1.0 - lr * weight_decay)
+            m.mul_(beta1).add_(gf, alpha=1.0 - beta1)
+            v.mul_(beta2).addcmul_(gf, gf, value=1.0 - beta2)
+            denom = (v / bc2).sqrt_().add_(eps)
+            w.addcdiv_(m, denom, value=-lr / bc1)
+            if mw is not None:
+                p.copy_(w.to(p.dtype))
+
+
+class FusedAdamW(torch.optim.Optimizer):
+    def __init__(self, params, lr=3e-5, betas=(0.9, 0.999), eps=1e-8,
+                 weight_decay=0.01, master_weights: bool = True):
+        defaults = dict(lr=lr, betas=betas, eps=eps, weight_decay=weight_decay)
+        super().__init__(params, defaults)
+        self.master_weights = master_weights
+
+    @torch.no_grad()
+    def step(self, closure=None, grad_scale_inv: float = 1.0):
+        loss = None
+        if closure is not None:
+            with torch.enable_grad():
+                loss = closure()
+        for group in self.param_groups:
+            params, grads, ms, vs, masters = [], [], [], [], []
+            for p in group["params"]:
+                if p.grad is None:
+                    continue
+                state = self.state[p]
+                if len(state) == 0:
+                    state["step"] = 0
+                    dev = p.device
+                    state["exp_avg"] = torch.zeros_like(p, dtype=torch.float32, device=dev)
+                    state["exp_avg_sq"] = torch.zeros_like(p, dtype=torch.float32, device=dev)
+                    if self.master_weights and p.dtype in (torch.bfloat16, torch.float16):
+                        state["master"] = p.detach().float().clone()
+                    else:
+                        state["master"] = None
+                state["step"] += 1
+                params.append(p)
+                grads.append(p.grad)
+                ms.append(state["exp_avg"])
+                vs.append(state["exp_avg_sq"])
+                masters.append(state["master"])
+            if params:
+                step = self.state[params[0]]["step"]
+                b1, b2 = group["betas"]
+                multi_tensor_adamw(params, grads, ms, vs, masters,
+                                   group["lr"], b1, b2, group["eps"],
+                                   group["weight_decay"], step, grad_scale_inv)
+        return loss
+
+
+class FusedSGD(torch.optim.Optimizer):
+    """SGD with momentum (K15 — fabric alt-optimizer path,
+    reference: fabric/fabric-cls.py:283-285)."""
+
+    def __init__(self, params, lr=1e-3, momentum=0.9, weight_decay=0.0):
+        super().__init__(params, dict(lr=lr, momentum=momentum,
+                                      weight_decay=weight_decay))
+
+    @torch.no_grad()
+    def step(self, closure=None):
+        loss = None
+        if closure is not None:
+            with torch.enable_grad():
+                loss = closure()
+        for group in self.param_groups:
+            for p in group["params"]:
+                if p.grad is None:
+                    continue
+                state = self.state[p]
+                g = p.grad.float()
+                if group["weight_decay"]:
+                    g = g.add(p.float(), alpha=group["weight_decay"])
+                if group["momentum"]:
+                    if "momentum_buffer" not in state:
+                        state["momentum_buffer"] = g.clone()
+                    else:
+                        state["momentum_buffer"].mul_(group["momentum"]).add_(g)
+                    g = state["momentum_buffer"]
+                p.add_((g * -group["lr"]).to(p.dtype))
+        return loss
+
+
+NO_DECAY_MARKERS = ("bias", "LayerNorm.weight", "layer_norm.weight", "ln.weight")
+
+
+def build_optimizer(model: torch.nn.Module, lr=3e-5, weight_decay=0.01,
+                    betas=(0.9, 0.999), eps=1e-8, optimizer="adamw",
+                    sgd_momentum=0.9, master_weights=True):
+    """Two param groups with the reference's no-decay rule
+    (single-gpu-cls.py:86-97)."""
+    decay, no_decay = [], []
+    for n, p in model.named_parameters():
+        if not p.requires_grad:
+            continue
+        (no_decay if any(m in n for m in NO_DECAY_MARKERS) else decay).append(p)
+    groups = [
+        {"params": decay, "weight_decay": weight_decay},
+        {"params": no_decay, "weight_decay": 0.0},
+    ]
+    if optimizer == "sgd":
+        return FusedSGD(groups, lr=lr, momentum=sgd_momentum)
+    return FusedAdamW(groups, lr=lr, betas=betas, eps=eps,
+                      weight_decay=weight_decay, master_weights=master_weights)

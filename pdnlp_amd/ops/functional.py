@@ -1,0 +1,302 @@
+"""Functional op API with HIP/torch dispatch.
+
+Each public function runs the hand-written CDNA4 kernel when its input lives
+on a ROCm GPU and the in-tree extension is built, and a plain differentiable
+torch composition otherwise (the numerics reference the GPU kernels are
+tested against). Kernel inventory mirrors SURVEY.md §2.3 K1-K16.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Optional
+
+import torch
+import torch.nn.functional as F
+
+from . import hip_enabled, ext
+
+
+# --------------------------------------------------------------------------
+# LayerNorm (K1 partner; fp32 stats, bf16/fp16/fp32 in/out)
+# --------------------------------------------------------------------------
+
+class _LayerNormFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, eps):
+        x = x.contiguous()
+        y, mean, rstd = ext().layernorm_fwd(x, weight, bias, eps)
+        ctx.save_for_backward(x, weight, mean, rstd)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight, mean, rstd = ctx.saved_tensors
+        dx, dw, db = ext().layernorm_bwd(dy.contiguous(), x, weight, mean, rstd)
+        return dx, dw, db, None
+
+
+def layernorm(x: torch.Tensor, weight: torch.Tensor, bias: torch.Tensor,
+              eps: float = 1e-12) -> torch.Tensor:
+    if hip_enabled(x):
+        return _LayerNormFn.apply(x, weight, bias, eps)
+    return F.layer_norm(x, (x.shape[-1],), weight, bias, eps)
+
+
+# --------------------------------------------------------------------------
+# Embedding gather + add + LayerNorm (K1): word+pos+type embed fused with LN
+# --------------------------------------------------------------------------
+
+class _EmbeddingLNFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, input_ids, token_type_ids, position_ids,
+                word_w, pos_w, type_w, ln_w, ln_b, eps):
+        y, mean, rstd = ext().embedding_ln_fwd(
+            input_ids, token_type_ids, position_ids,
+            word_w, pos_w, type_w, ln_w, ln_b, eps)
+        ctx.save_for_backward(input_ids, token_type_ids, position_ids,
+                              word_w, pos_w, type_w, ln_w, mean, rstd)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (ids, type_ids, pos_ids, word_w, pos_w, type_w, ln_w,
+         mean, rstd) = ctx.saved_tensors
+        dword, dpos, dtype_, dlnw, dlnb = ext().embedding_ln_bwd(
+            dy.contiguous(), ids, type_ids, pos_ids,
+            word_w, pos_w, type_w, ln_w, mean, rstd)
+        return (None, None, None, dword, dpos, dtype_, dlnw, dlnb, None)
+
+
+def embedding_layernorm(input_ids, token_type_ids, position_ids,
+                        word_w, pos_w, type_w, ln_w, ln_b,
+                        eps: float = 1e-12) -> torch.Tensor:
+    if hip_enabled(word_w):
+        return _EmbeddingLNFn.apply(
+            input_ids.contiguous(), token_type_ids.contiguous(),
+            position_ids.contiguous(), word_w, pos_w, type_w, ln_w, ln_b, eps)
+    emb = (F.embedding(input_ids, word_w)
+           + F.embedding(position_ids, pos_w)
+           + F.embedding(token_type_ids, type_w))
+    return F.layer_norm(emb, (emb.shape[-1],), ln_w, ln_b, eps)
+
+
+# --------------------------------------------------------------------------
+# Linear (K2/K6/K7/K8 GEMMs). HIP MFMA GEMM for the hot shapes; the
+# backward dGEMMs (plain, no fusion) go through rocBLAS via torch.matmul.
+# --------------------------------------------------------------------------
+
+class _LinearHipFn(torch.autograd.Function):
+    """y = x @ w^T + b with the forward GEMM on the hand-written MFMA kernel
+    (optionally fused activation), backward dGEMMs on rocBLAS."""
+
+    @staticmethod
+    def forward(ctx, x, w, b, act):
+        x2 = x.contiguous().view(-1, x.shape[-1])
+        y, pre = ext().gemm_nt_fwd(x2, w, b if b is not None else torch.Tensor(), act)
+        ctx.save_for_backward(x2, w, pre if act != "none" else torch.Tensor())
+        ctx.act = act
+        ctx.has_bias = b is not None
+        ctx.in_shape = x.shape
+        return y.view(*x.shape[:-1], w.shape[0])
+
+    @staticmethod
+    def backward(ctx, dy):
+        x2, w, pre = ctx.saved_tensors
+        dy2 = dy.contiguous().view(-1, dy.shape[-1])
+        if ctx.act == "gelu":
+            dy2 = ext().gelu_bwd(dy2, pre)
+        elif ctx.act == "tanh":
+            dy2 = ext().tanh_bwd(dy2, pre)
+        dx = dy2 @ w                      # rocBLAS NN
+        dw = dy2.t() @ x2                 # rocBLAS TN
+        db = dy2.sum(0) if ctx.has_bias else None
+        return dx.view(ctx.in_shape), dw, db, None
+
+
+def linear(x: torch.Tensor, w: torch.Tensor, b: Optional[torch.Tensor] = None,
+           act: str = "none") -> torch.Tensor:
+    """act in {"none", "gelu", "tanh"} — fused into the GEMM epilogue on HIP."""
+    if hip_enabled(x) and getattr(ext(), "gemm_nt_fwd", None) is not None \
+            and _gemm_shape_ok(x, w):
+        return _LinearHipFn.apply(x, w, b, act)
+    y = F.linear(x, w, b)
+    if act == "gelu":
+        y = F.gelu(y)
+    elif act == "tanh":
+        y = torch.tanh(y)
+    return y
+
+
+def _gemm_shape_ok(x, w) -> bool:
+    if x.dtype not in (torch.bfloat16, torch.float16):
+        return False
+    m = x.numel() // x.shape[-1]
+    n, k = w.shape
+    return (k % 64 == 0) and (n % 64 == 0) and (m % 16 == 0)
+
+
+# --------------------------------------------------------------------------
+# Bias + GELU (K7 epilogue when unfused)
+# --------------------------------------------------------------------------
+
+class _BiasGeluFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, b):
+        y = ext().bias_gelu_fwd(x.contiguous(), b)
+        ctx.save_for_backward(x, b)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, b = ctx.saved_tensors
+        dx, db = ext().bias_gelu_bwd(dy.contiguous(), x, b)
+        return dx, db
+
+
+def bias_gelu(x: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+    if hip_enabled(x):
+        return _BiasGeluFn.apply(x, b)
+    return F.gelu(x + b)
+
+
+# --------------------------------------------------------------------------
+# Masked softmax (K4): y = softmax(scores * scale + additive_mask)
+# --------------------------------------------------------------------------
+
+class _MaskedSoftmaxFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, scores, mask, scale):
+        probs = ext().masked_softmax_fwd(scores.contiguous(),
+                                         mask.contiguous() if mask is not None
+                                         else torch.Tensor(), scale)
+        ctx.save_for_backward(probs)
+        return probs
+
+    @staticmethod
+    def backward(ctx, dy):
+        (probs,) = ctx.saved_tensors
+        return ext().masked_softmax_bwd(dy.contiguous(), probs), None, None
+
+
+def masked_softmax(scores: torch.Tensor, mask: Optional[torch.Tensor],
+                   scale: float = 1.0) -> torch.Tensor:
+    """scores: [B, H, S, S]; mask: additive [B, 1, 1, S] (0 keep / -inf drop)."""
+    if hip_enabled(scores):
+        return _MaskedSoftmaxFn.apply(scores, mask, scale)
+    s = scores * scale
+    if mask is not None:
+        s = s + mask
+    return F.softmax(s, dim=-1)
+
+
+# --------------------------------------------------------------------------
+# Attention (K3-K5). Default: batched GEMMs + fused masked softmax.
+# Flash path (fused MFMA online-softmax kernel) used when available.
+# --------------------------------------------------------------------------
+
+def attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+              mask: Optional[torch.Tensor], dropout_p: float = 0.0,
+              training: bool = False) -> torch.Tensor:
+    """q,k,v: [B, H, S, D]; mask: additive [B, 1, 1, S]. Returns [B, H, S, D]."""
+    scale = 1.0 / math.sqrt(q.shape[-1])
+    if hip_enabled(q) and getattr(ext(), "flash_attn_fwd", None) is not None \
+            and (dropout_p == 0.0 or not training) and q.shape[-1] in (64,) \
+            and q.dtype == torch.bfloat16:
+        return _FlashAttnFn.apply(q.contiguous(), k.contiguous(), v.contiguous(),
+                                  mask, scale)
+    scores = torch.matmul(q, k.transpose(-1, -2))
+    probs = masked_softmax(scores, mask, scale)
+    if dropout_p > 0.0 and training:
+        probs = F.dropout(probs, p=dropout_p, training=True)
+    return torch.matmul(probs, v)
+
+
+class _FlashAttnFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, mask, scale):
+        o, lse = ext().flash_attn_fwd(q, k, v,
+                                      mask.contiguous() if mask is not None
+                                      else torch.Tensor(), scale)
+        ctx.save_for_backward(q, k, v, o, lse,
+                              mask if mask is not None else torch.Tensor())
+        ctx.scale = scale
+        return o
+
+    @staticmethod
+    def backward(ctx, do):
+        q, k, v, o, lse, mask = ctx.saved_tensors
+        dq, dk, dv = ext().flash_attn_bwd(do.contiguous(), q, k, v, o, lse,
+                                          mask, ctx.scale)
+        return dq, dk, dv, None, None
+
+
+# --------------------------------------------------------------------------
+# Bias + dropout + residual + LayerNorm (K6/K8 epilogue)
+# --------------------------------------------------------------------------
+
+class _BiasDropResLNFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, y, bias, residual, ln_w, ln_b, p, training, eps, seed):
+        out, xsum, mask, mean, rstd = ext().bias_dropout_residual_ln_fwd(
+            y.contiguous(), bias, residual.contiguous(), ln_w, ln_b,
+            float(p if training else 0.0), eps, int(seed))
+        ctx.save_for_backward(xsum, mask, ln_w, mean, rstd)
+        ctx.p = float(p if training else 0.0)
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        xsum, mask, ln_w, mean, rstd = ctx.saved_tensors
+        dy, dbias, dres, dlnw, dlnb = ext().bias_dropout_residual_ln_bwd(
+            dout.contiguous(), xsum, mask, ln_w, mean, rstd, ctx.p)
+        return dy, dbias, dres, dlnw, dlnb, None, None, None, None
+
+
+def bias_dropout_residual_layernorm(
+        y: torch.Tensor, bias: torch.Tensor, residual: torch.Tensor,
+        ln_w: torch.Tensor, ln_b: torch.Tensor, p: float = 0.1,
+        training: bool = False, eps: float = 1e-12) -> torch.Tensor:
+    """out = LN(dropout(y + bias) + residual) — the fused epilogue after the
+    attention-output and FFN-down projections."""
+    if hip_enabled(y):
+        seed = int(torch.randint(0, 2**31 - 1, (1,)).item()) if (p > 0 and training) else 0
+        return _BiasDropResLNFn.apply(y, bias, residual, ln_w, ln_b, p,
+                                      training, eps, seed)
+    h = y + bias
+    if p > 0.0 and training:
+        h = F.dropout(h, p=p, training=True)
+    h = h + residual
+    return F.layer_norm(h, (h.shape[-1],), ln_w, ln_b, eps)
+
+
+# --------------------------------------------------------------------------
+# Dropout (K16) — torch native RNG kernel on both paths for now
+# --------------------------------------------------------------------------
+
+def dropout(x: torch.Tensor, p: float, training: bool) -> torch.Tensor:
+    return F.dropout(x, p=p, training=training)
+
+
+# --------------------------------------------------------------------------
+# CrossEntropy (K10): fused log-softmax + NLL on [B, num_labels]
+# --------------------------------------------------------------------------
+
+class _CrossEntropyFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, logits, labels):
+        loss, logprobs = ext().cross_entropy_fwd(logits.contiguous(),
+                                                 labels.contiguous())
+        ctx.save_for_backward(logprobs, labels)
+        return loss
+
+    @staticmethod
+    def backward(ctx, dloss):
+        logprobs, labels = ctx.saved_tensors
+        return ext().cross_entropy_bwd(dloss, logprobs, labels), None
+
+
+def cross_entropy(logits: torch.Tensor, labels: torch.Tensor) -> torch.Tensor:
+    if hip_enabled(logits) and getattr(ext(), "cross_entropy_fwd", None) is not None:
+        return _CrossEntropyFn.apply(logits.float(), labels)
+    return F.cross_entropy(logits.float(), labels)

@@ -1,0 +1,109 @@
+"""Process bootstrap & topology (layer L6 of SURVEY.md §1).
+
+Covers both reference launch modes:
+- external launcher (torchrun / torch.distributed.launch): env-var rendezvous
+  MASTER_ADDR/PORT/RANK/WORLD_SIZE/LOCAL_RANK (reference:
+  multi-gpu-distributed-cls.py:275-284)
+- self-spawn: ``spawn(worker, nprocs)`` with a TCP store on 127.0.0.1
+  (reference: multi-gpu-distributed-mp-cls.py:265-266, :361)
+
+Backend "nccl" IS RCCL on ROCm; "gloo" is the CPU test path.
+"""
+
+from __future__ import annotations
+
+import datetime
+import os
+from typing import Callable, Optional
+
+import torch
+import torch.distributed as dist
+
+
+def _default_backend() -> str:
+    return "nccl" if torch.cuda.is_available() else "gloo"
+
+
+def init_distributed(backend: Optional[str] = None,
+                     init_method: Optional[str] = None,
+                     world_size: Optional[int] = None,
+                     rank: Optional[int] = None,
+                     timeout_sec: int = 600) -> int:
+    """Initialize the process group and bind the device.
+
+    Returns local_rank. A world_size of 1 with no launcher env is the
+    degenerate single-process mode (no process group is created).
+    """
+    env_ws = int(os.environ.get("WORLD_SIZE", world_size or 1))
+    if env_ws <= 1 and world_size in (None, 1):
+        local = int(os.environ.get("LOCAL_RANK", 0))
+        if torch.cuda.is_available():
+            torch.cuda.set_device(local)
+        return local
+
+    backend = backend or _default_backend()
+    kwargs = dict(backend=backend,
+                  timeout=datetime.timedelta(seconds=timeout_sec))
+    if init_method is not None:
+        kwargs.update(init_method=init_method,
+                      world_size=world_size, rank=rank)
+    else:
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29500")
+        if world_size is not None:
+            kwargs.update(world_size=world_size)
+        if rank is not None:
+            kwargs.update(rank=rank)
+    if not dist.is_initialized():
+        dist.init_process_group(**kwargs)
+    local = int(os.environ.get("LOCAL_RANK",
+                               rank if rank is not None else dist.get_rank()))
+    if torch.cuda.is_available():
+        torch.cuda.set_device(local % max(torch.cuda.device_count(), 1))
+    return local
+
+
+def cleanup() -> None:
+    if dist.is_available() and dist.is_initialized():
+        dist.destroy_process_group()
+
+
+def spawn(worker: Callable, nprocs: int, args: tuple = (),
+          master_port: int = 12355, backend: Optional[str] = None) -> None:
+    """mp.spawn launch mode: forks nprocs workers, each calls
+    ``worker(local_rank, nprocs, *args)`` after we set the env contract."""
+    import torch.multiprocessing as mp
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(master_port)
+    os.environ["WORLD_SIZE"] = str(nprocs)
+    mp.spawn(_spawn_entry, nprocs=nprocs,
+             args=(worker, nprocs, backend, args), join=True)
+
+
+def _spawn_entry(local_rank: int, worker: Callable, nprocs: int,
+                 backend: Optional[str], args: tuple) -> None:
+    os.environ["RANK"] = str(local_rank)
+    os.environ["LOCAL_RANK"] = str(local_rank)
+    init_distributed(backend=backend, world_size=nprocs, rank=local_rank)
+    try:
+        worker(local_rank, nprocs, *args)
+    finally:
+        cleanup()
+
+
+def get_rank() -> int:
+    return dist.get_rank() if dist.is_initialized() else 0
+
+
+def get_world_size() -> int:
+    return dist.get_world_size() if dist.is_initialized() else 1
+
+
+def get_local_rank() -> int:
+    return int(os.environ.get("LOCAL_RANK", 0))
+
+
+def barrier() -> None:
+    if dist.is_initialized():
+        dist.barrier()

@@ -1,0 +1,232 @@
+"""ZeRO-style sharded data parallelism (the DeepSpeed capability).
+
+Reference: DeepSpeed ZeRO-3 engine — grad reduce-scatter, sharded AdamW with
+fp32 master weights, param allgather, sharded checkpoints + ``zero_to_fp32``
+consolidation (multi-gpu-deepspeed-cls.py:220-247, README.md:484-488).
+
+MI355X-native design: with 288 GB HBM3E per GPU, parameter partitioning with
+per-layer prefetch (ZeRO-3's main point on 40-80 GB cards) buys nothing at
+BERT scale — the observable capabilities (memory drop from sharded optimizer
+state + master weights, sharded checkpoints, consolidation tool) come from a
+ZeRO-1/2 design: gradients reduce-scattered over xGMI so each rank reduces
+1/N of the volume, AdamW runs on the local shard only (fused multi-tensor
+kernel), updated params all-gather back. Collectives are chunked at
+``bucket_mb`` and launched on the compute stream at step() (the grads must
+all exist), which on RCCL rings keeps every link busy with 1/N the bytes of
+plain DDP all-reduce.
+"""
+
+from __future__ import annotations
+
+import os
+from typing import Dict, List, Optional
+
+import torch
+import torch.distributed as dist
+
+from ..ops.adamw import NO_DECAY_MARKERS, multi_tensor_adamw
+
+
+class _FlatGroup:
+    """One weight-decay group flattened into param/grad flats, shard-split."""
+    __slots__ = ("names", "params", "shapes", "offsets", "numel_padded",
+                 "param_flat", "grad_flat", "master_shard", "m_shard",
+                 "v_shard", "weight_decay", "shard_size")
+
+    def __init__(self):
+        self.names: List[str] = []
+        self.params: List[torch.nn.Parameter] = []
+        self.shapes = []
+        self.offsets = []
+
+
+class ZeroRedundancyOptimizer:
+    """Sharded AdamW over flat parameter groups.
+
+    Use: build model (on device, final dtype), then
+    ``opt = ZeroRedundancyOptimizer(model, ...)``; training loop is
+    ``loss.backward(); opt.step(); opt.zero_grad()``. Parameter broadcast at
+    construction keeps ranks identical (DDP-equivalent init).
+    """
+
+    def __init__(self, model: torch.nn.Module, lr: float = 3e-5,
+                 betas=(0.9, 0.999), eps: float = 1e-8,
+                 weight_decay: float = 0.01, bucket_mb: float = 200.0,
+                 group=None):
+        self.model = model
+        self.lr = lr
+        self.betas = betas
+        self.eps = eps
+        self.group = group
+        self.world = dist.get_world_size(group) if dist.is_initialized() else 1
+        self.rank = dist.get_rank(group) if dist.is_initialized() else 0
+        self.step_count = 0
+        self.bucket_bytes = int(bucket_mb * 1024 * 1024)
+
+        if dist.is_initialized():
+            for t in model.state_dict().values():
+                if isinstance(t, torch.Tensor) and t.numel() > 0:
+                    dist.broadcast(t.data, src=0, group=group)
+
+        self.groups: List[_FlatGroup] = []
+        decay, no_decay = _FlatGroup(), _FlatGroup()
+        decay.weight_decay = weight_decay
+        no_decay.weight_decay = 0.0
+        for n, p in model.named_parameters():
+            if not p.requires_grad:
+                continue
+            g = no_decay if any(m in n for m in NO_DECAY_MARKERS) else decay
+            g.names.append(n)
+            g.params.append(p)
+            g.shapes.append(p.shape)
+        for g in (decay, no_decay):
+            if g.params:
+                self._flatten_group(g)
+                self.groups.append(g)
+
+    def _flatten_group(self, g: _FlatGroup):
+        dev = g.params[0].device
+        dtype = g.params[0].dtype
+        total = sum(p.numel() for p in g.params)
+        g.numel_padded = ((total + self.world - 1) // self.world) * self.world
+        g.shard_size = g.numel_padded // self.world
+        g.param_flat = torch.zeros(g.numel_padded, dtype=dtype, device=dev)
+        g.grad_flat = torch.zeros(g.numel_padded, dtype=dtype, device=dev)
+        off = 0
+        g.offsets = []
+        for p in g.params:
+            n = p.numel()
+            g.offsets.append(off)
+            g.param_flat[off:off + n].copy_(p.data.reshape(-1))
+            p.data = g.param_flat[off:off + n].view_as(p)
+            p.grad = g.grad_flat[off:off + n].view_as(p)
+            off += n
+        lo = self.rank * g.shard_size
+        hi = lo + g.shard_size
+        g.master_shard = g.param_flat[lo:hi].float().clone()
+        g.m_shard = torch.zeros_like(g.master_shard)
+        g.v_shard = torch.zeros_like(g.master_shard)
+
+    # ------------------------------------------------------------------
+    @torch.no_grad()
+    def step(self, closure=None, grad_scale_inv: float = 1.0):
+        self.step_count += 1
+        for g in self.groups:
+            lo = self.rank * g.shard_size
+            shard_grad = self._reduce_scatter(g)
+            multi_tensor_adamw(
+                [g.param_flat[lo:lo + g.shard_size]], [shard_grad],
+                [g.m_shard], [g.v_shard], [g.master_shard],
+                self.lr, self.betas[0], self.betas[1], self.eps,
+                g.weight_decay, self.step_count, grad_scale_inv)
+            self._all_gather(g)
+
+    def _reduce_scatter(self, g: _FlatGroup) -> torch.Tensor:
+        lo = self.rank * g.shard_size
+        shard = g.grad_flat[lo:lo + g.shard_size]
+        if not dist.is_initialized() or self.world == 1:
+            return shard
+        backend = dist.get_backend(self.group)
+        if backend == "nccl":
+            out = torch.empty_like(shard)
+            # chunked so a bucket_mb knob bounds peak wire message size
+            n = g.shard_size
+            step = max(self.bucket_bytes // shard.element_size(), 1)
+            for s in range(0, n, step):
+                e = min(s + step, n)
+                seg_in = g.grad_flat.view(self.world, n)[:, s:e].contiguous()
+                dist.reduce_scatter_tensor(out[s:e], seg_in, group=self.group)
+            out.div_(self.world)
+            return out
+        # gloo fallback: all_reduce then slice (functionally identical)
+        dist.all_reduce(g.grad_flat, group=self.group)
+        g.grad_flat.div_(self.world)
+        return shard
+
+    def _all_gather(self, g: _FlatGroup):
+        if not dist.is_initialized() or self.world == 1:
+            return
+        backend = dist.get_backend(self.group)
+        lo = self.rank * g.shard_size
+        if backend == "nccl":
+            dist.all_gather_into_tensor(
+                g.param_flat, g.param_flat[lo:lo + g.shard_size].contiguous(),
+                group=self.group)
+        else:
+            chunks = list(g.param_flat.view(self.world, g.shard_size).unbind(0))
+            dist.all_gather(chunks,
+                            g.param_flat[lo:lo + g.shard_size].clone(),
+                            group=self.group)
+            for i, c in enumerate(chunks):
+                g.param_flat.view(self.world, g.shard_size)[i].copy_(c)
+
+    def zero_grad(self, set_to_none: bool = False):
+        for g in self.groups:
+            g.grad_flat.zero_()
+
+    # ---- sharded checkpointing (deepspeed save_checkpoint equivalent) ----
+    def save_checkpoint(self, ckpt_dir: str, tag: str = "latest"):
+        os.makedirs(ckpt_dir, exist_ok=True)
+        payload = {
+            "world": self.world, "rank": self.rank,
+            "step": self.step_count,
+            "groups": [{
+                "names": g.names,
+                "shapes": [list(s) for s in g.shapes],
+                "offsets": g.offsets,
+                "numel_padded": g.numel_padded,
+                "shard_size": g.shard_size,
+                "weight_decay": g.weight_decay,
+                "master_shard": g.master_shard.cpu(),
+                "m_shard": g.m_shard.cpu(),
+                "v_shard": g.v_shard.cpu(),
+            } for g in self.groups],
+        }
+        torch.save(payload, os.path.join(
+            ckpt_dir, f"zero_shard_r{self.rank:02d}.pt"))
+        if self.rank == 0:
+            with open(os.path.join(ckpt_dir, tag), "w") as f:
+                f.write("zero checkpoint")
+
+    def load_checkpoint(self, ckpt_dir: str):
+        path = os.path.join(ckpt_dir, f"zero_shard_r{self.rank:02d}.pt")
+        payload = torch.load(path, map_location="cpu", weights_only=False)
+        if payload["world"] != self.world:
+            raise RuntimeError("world size mismatch on zero checkpoint load")
+        self.step_count = payload["step"]
+        for g, saved in zip(self.groups, payload["groups"]):
+            g.master_shard.copy_(saved["master_shard"].to(g.master_shard.device))
+            g.m_shard.copy_(saved["m_shard"].to(g.m_shard.device))
+            g.v_shard.copy_(saved["v_shard"].to(g.v_shard.device))
+            lo = self.rank * g.shard_size
+            g.param_flat[lo:lo + g.shard_size].copy_(
+                g.master_shard.to(g.param_flat.dtype))
+        for g in self.groups:
+            self._all_gather(g)
+
+
+def consolidate_zero_checkpoint(ckpt_dir: str, out_path: Optional[str] = None
+                                ) -> Dict[str, torch.Tensor]:
+    """``zero_to_fp32``-equivalent: merge per-rank shards into a full fp32
+    model state dict (reference: README.md:484-488)."""
+    import glob
+
+    shards = sorted(glob.glob(os.path.join(ckpt_dir, "zero_shard_r*.pt")))
+    if not shards:
+        raise FileNotFoundError(f"no zero shards in {ckpt_dir}")
+    payloads = [torch.load(s, map_location="cpu", weights_only=False)
+                for s in shards]
+    world = payloads[0]["world"]
+    assert len(payloads) == world, "missing shards"
+    sd: Dict[str, torch.Tensor] = {}
+    for gi, meta in enumerate(payloads[0]["groups"]):
+        flat = torch.cat([p["groups"][gi]["master_shard"] for p in payloads])
+        for name, shape, off in zip(meta["names"], meta["shapes"],
+                                    meta["offsets"]):
+            n = 1
+            for s in shape:
+                n *= s
+            sd[name] = flat[off:off + n].view(*shape).clone()
+    if out_path:
+        torch.save(sd, out_path)
+    return sd

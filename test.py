@@ -1,0 +1,59 @@
+#!/usr/bin/env python
+"""Offline checkpoint audit (reference: test.py).
+
+Loads each strategy's saved checkpoint on one device (bare model — our
+checkpoints are saved unwrapped; reference-style ``module.``-prefixed dicts
+are stripped automatically) and scores the dev split with a classification
+report. ZeRO sharded checkpoint dirs are consolidated on the fly.
+
+    python test.py [--ckpt output/model.pt ...]
+"""
+import argparse
+import glob
+import os
+
+import torch
+
+from pdnlp_amd.config import Args
+from pdnlp_amd.cli import build_dataloaders
+from pdnlp_amd.data import LABELS
+from pdnlp_amd.engine.trainer import Trainer
+from pdnlp_amd.models import build_model
+from pdnlp_amd.ops.adamw import build_optimizer
+from pdnlp_amd.parallel.zero import consolidate_zero_checkpoint
+from pdnlp_amd.utils import set_seed, load_checkpoint, strip_module_prefix
+
+
+def evaluate_checkpoint(path: str, args: Args):
+    device = torch.device("cuda:0" if torch.cuda.is_available() else "cpu")
+    model = build_model("bert-base", model_path=args.model_path)
+    if os.path.isdir(path) and glob.glob(os.path.join(path, "zero_shard_r*.pt")):
+        sd = consolidate_zero_checkpoint(path)
+        model.load_state_dict({k: v for k, v in sd.items()}, strict=False)
+    else:
+        load_checkpoint(model, path)
+    model = model.to(device)
+    _, dev_loader, _ = build_dataloaders(args, 1, 0)
+    trainer = Trainer(args, model, build_optimizer(model), device)
+    loss, acc, report = trainer.test(dev_loader, label_names=LABELS)
+    print(f"== {path}: dev loss {loss:.4f} acc {acc:.4f}")
+    return acc
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--ckpt", action="append", default=None,
+                   help="checkpoint file/dir; repeatable. Default: output/*.pt")
+    ns, rest = p.parse_known_args()
+    args = Args().apply_cli(rest)
+    set_seed(args.seed)
+    ckpts = ns.ckpt or sorted(glob.glob(os.path.join(args.output_dir, "*.pt")))
+    if not ckpts:
+        print(f"no checkpoints found under {args.output_dir}")
+        return
+    for c in ckpts:
+        evaluate_checkpoint(c, args)
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,34 @@
+import os
+import sys
+
+import pytest
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
+def pytest_configure(config):
+    config.addinivalue_line(
+        "markers", "gpu: test requires a ROCm GPU (run on an MI355X box)")
+
+
+def pytest_collection_modifyitems(config, items):
+    if torch.cuda.is_available():
+        return
+    skip = pytest.mark.skip(reason="no GPU in this environment")
+    for item in items:
+        if "gpu" in item.keywords:
+            item.add_marker(skip)
+
+
+@pytest.fixture
+def tiny_cfg():
+    from pdnlp_amd.config import BertConfig
+    return BertConfig.tiny()
+
+
+@pytest.fixture(autouse=True)
+def _seed_everything():
+    from pdnlp_amd.utils import set_seed
+    set_seed(123)
+    yield

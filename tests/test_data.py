@@ -1,0 +1,59 @@
+import json
+
+import torch
+
+from pdnlp_amd.data import (CharTokenizer, ClsDataset, Collate,
+                            DistributedSampler, SyntheticClsDataset,
+                            load_data, train_dev_split)
+
+
+def test_load_data_strips_spaces(tmp_path):
+    p = tmp_path / "train.json"
+    p.write_text(json.dumps([["你 好 世 界", 2], ["a b c", 5]]),
+                 encoding="utf-8")
+    data = load_data(str(p))
+    assert data == [("你好世界", 2), ("abc", 5)]
+
+
+def test_split_deterministic():
+    data = [(f"t{i}", i % 6) for i in range(100)]
+    a1, b1 = train_dev_split(data, 0.92, seed=123)
+    a2, b2 = train_dev_split(data, 0.92, seed=123)
+    assert a1 == a2 and b1 == b2
+    assert len(a1) == 92 and len(b1) == 8
+
+
+def test_collate_shapes():
+    tok = CharTokenizer(vocab_size=512)
+    collate = Collate(tok, max_seq_len=32)
+    batch = collate([("你好世界", 1), ("测试", 4)])
+    assert batch["input_ids"].shape == (2, 32)
+    assert batch["attention_mask"].shape == (2, 32)
+    assert batch["token_type_ids"].shape == (2, 32)
+    assert batch["label"].tolist() == [1, 4]
+    assert batch["input_ids"][0, 0].item() == 101  # [CLS]
+    assert batch["attention_mask"][0].sum().item() == 6  # CLS + 4 chars + SEP
+
+
+def test_synthetic_dataset_deterministic():
+    ds = SyntheticClsDataset(10, seq_len=16, vocab_size=512)
+    a, b = ds[3], ds[3]
+    assert torch.equal(a["input_ids"], b["input_ids"])
+    assert a["input_ids"].shape == (16,)
+
+
+def test_distributed_sampler_sharding():
+    ds = ClsDataset([(f"t{i}", 0) for i in range(10)])
+    s0 = DistributedSampler(ds, num_replicas=3, rank=0, shuffle=False)
+    s1 = DistributedSampler(ds, num_replicas=3, rank=1, shuffle=False)
+    s2 = DistributedSampler(ds, num_replicas=3, rank=2, shuffle=False)
+    i0, i1, i2 = list(s0), list(s1), list(s2)
+    assert len(i0) == len(i1) == len(i2) == 4  # padded to 12
+    assert len(set(i0 + i1 + i2)) == 10  # covers all samples
+    # shuffling changes with epoch
+    s0 = DistributedSampler(ds, num_replicas=2, rank=0, shuffle=True, seed=1)
+    s0.set_epoch(0)
+    e0 = list(s0)
+    s0.set_epoch(1)
+    e1 = list(s0)
+    assert e0 != e1
