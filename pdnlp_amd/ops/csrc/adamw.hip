@@ -1,0 +1,184 @@
+// Multi-tensor AdamW update + AMP unscale/non-finite check
+// (SURVEY.md K12/K13).
+//
+// Chunked multi-tensor launch: up to 32 tensors per launch packed into the
+// kernel-argument block (pointer arrays + per-tensor block-offset prefix
+// sums; the kernel finds its tensor by scanning the 33-entry prefix array —
+// trivially cheap on the scalar unit). Replaces torch's per-tensor foreach
+// launches with one launch per ~32 tensors: the reference's AdamW walks 201
+// BERT tensors per step.
+//
+// Honors decoupled weight decay (per group), fp32 m/v, optional fp32 master
+// weights for bf16/fp16 params, and a fused grad-scale-inverse (AMP).
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "common.h"
+
+namespace {
+
+constexpr int MAXT = 32;
+constexpr int BLOCK = 256;
+constexpr int ILP = 4;
+
+struct AdamWMeta {
+  const void* g[MAXT];
+  void* p[MAXT];
+  float* m[MAXT];
+  float* v[MAXT];
+  float* mw[MAXT];  // nullptr when no master weights
+  long numel[MAXT];
+  int block_prefix[MAXT + 1];
+  int ntensors;
+};
+
+template <typename T, bool MASTER>
+__global__ void adamw_kernel(AdamWMeta meta, float lr, float beta1,
+                             float beta2, float eps, float wd, float bc1,
+                             float bc2, float grad_scale_inv) {
+  // find tensor for this block
+  int t = 0;
+  while (t + 1 < meta.ntensors && blockIdx.x >= meta.block_prefix[t + 1]) ++t;
+  const long local_block = blockIdx.x - meta.block_prefix[t];
+  const long n = meta.numel[t];
+  const T* g = (const T*)meta.g[t];
+  T* p = (T*)meta.p[t];
+  float* m = meta.m[t];
+  float* v = meta.v[t];
+  float* mw = MASTER ? meta.mw[t] : nullptr;
+
+  const float step_size = lr / bc1;
+  const float wd_factor = 1.f - lr * wd;
+  const long base = local_block * BLOCK * ILP + threadIdx.x;
+#pragma unroll
+  for (int j = 0; j < ILP; ++j) {
+    const long i = base + (long)j * BLOCK;
+    if (i < n) {
+      const float gf = to_f32<T>(g[i]) * grad_scale_inv;
+      float w = MASTER ? mw[i] : to_f32<T>(p[i]);
+      w *= wd_factor;
+      const float mi = beta1 * m[i] + (1.f - beta1) * gf;
+      const float vi = beta2 * v[i] + (1.f - beta2) * gf * gf;
+      m[i] = mi;
+      v[i] = vi;
+      w -= step_size * mi / (sqrtf(vi / bc2) + eps);
+      if (MASTER) mw[i] = w;
+      p[i] = from_f32<T>(w);
+    }
+  }
+}
+
+struct UnscaleMeta {
+  void* g[MAXT];
+  long numel[MAXT];
+  int block_prefix[MAXT + 1];
+  int ntensors;
+};
+
+template <typename T>
+__global__ void unscale_kernel(UnscaleMeta meta, float* found_inf,
+                               float inv_scale) {
+  int t = 0;
+  while (t + 1 < meta.ntensors && blockIdx.x >= meta.block_prefix[t + 1]) ++t;
+  const long local_block = blockIdx.x - meta.block_prefix[t];
+  const long n = meta.numel[t];
+  T* g = (T*)meta.g[t];
+  const long base = local_block * BLOCK * ILP + threadIdx.x;
+  bool bad = false;
+#pragma unroll
+  for (int j = 0; j < ILP; ++j) {
+    const long i = base + (long)j * BLOCK;
+    if (i < n) {
+      const float x = to_f32<T>(g[i]) * inv_scale;
+      if (!isfinite(x)) bad = true;
+      g[i] = from_f32<T>(x);
+    }
+  }
+  if (__any(bad) && (threadIdx.x & (WAVE - 1)) == 0) *found_inf = 1.f;
+}
+
+int blocks_of(long numel) {
+  return (int)((numel + (long)BLOCK * ILP - 1) / ((long)BLOCK * ILP));
+}
+
+}  // namespace
+
+void multi_tensor_adamw(std::vector<torch::Tensor> params,
+                        std::vector<torch::Tensor> grads,
+                        std::vector<torch::Tensor> exp_avgs,
+                        std::vector<torch::Tensor> exp_avg_sqs,
+                        std::vector<torch::Tensor> masters, double lr,
+                        double beta1, double beta2, double eps,
+                        double weight_decay, double bc1, double bc2,
+                        double grad_scale_inv) {
+  TORCH_CHECK(!params.empty());
+  const bool master = !masters.empty();
+  auto stream = at::hip::getCurrentHIPStream();
+  const auto dtype = params[0].scalar_type();
+  size_t i = 0;
+  while (i < params.size()) {
+    AdamWMeta meta{};
+    int nt = 0;
+    int blocks = 0;
+    while (i < params.size() && nt < MAXT) {
+      auto& p = params[i];
+      TORCH_CHECK(p.scalar_type() == dtype, "mixed dtypes in one chunk");
+      meta.p[nt] = p.data_ptr();
+      meta.g[nt] = grads[i].data_ptr();
+      meta.m[nt] = exp_avgs[i].data_ptr<float>();
+      meta.v[nt] = exp_avg_sqs[i].data_ptr<float>();
+      meta.mw[nt] = master ? masters[i].data_ptr<float>() : nullptr;
+      meta.numel[nt] = p.numel();
+      meta.block_prefix[nt] = blocks;
+      blocks += blocks_of(p.numel());
+      ++nt;
+      ++i;
+    }
+    meta.block_prefix[nt] = blocks;
+    meta.ntensors = nt;
+    DISPATCH_FLOAT_TYPES(dtype, "multi_tensor_adamw", [&] {
+      if (master) {
+        hipLaunchKernelGGL((adamw_kernel<scalar_t, true>), dim3(blocks),
+                           dim3(BLOCK), 0, stream, meta, (float)lr,
+                           (float)beta1, (float)beta2, (float)eps,
+                           (float)weight_decay, (float)bc1, (float)bc2,
+                           (float)grad_scale_inv);
+      } else {
+        hipLaunchKernelGGL((adamw_kernel<scalar_t, false>), dim3(blocks),
+                           dim3(BLOCK), 0, stream, meta, (float)lr,
+                           (float)beta1, (float)beta2, (float)eps,
+                           (float)weight_decay, (float)bc1, (float)bc2,
+                           (float)grad_scale_inv);
+      }
+    });
+  }
+}
+
+void multi_tensor_unscale(std::vector<torch::Tensor> grads,
+                          torch::Tensor found_inf, double inv_scale) {
+  TORCH_CHECK(!grads.empty());
+  auto stream = at::hip::getCurrentHIPStream();
+  const auto dtype = grads[0].scalar_type();
+  size_t i = 0;
+  while (i < grads.size()) {
+    UnscaleMeta meta{};
+    int nt = 0;
+    int blocks = 0;
+    while (i < grads.size() && nt < MAXT) {
+      meta.g[nt] = grads[i].data_ptr();
+      meta.numel[nt] = grads[i].numel();
+      meta.block_prefix[nt] = blocks;
+      blocks += blocks_of(grads[i].numel());
+      ++nt;
+      ++i;
+    }
+    meta.block_prefix[nt] = blocks;
+    meta.ntensors = nt;
+    DISPATCH_FLOAT_TYPES(dtype, "multi_tensor_unscale", [&] {
+      hipLaunchKernelGGL((unscale_kernel<scalar_t>), dim3(blocks), dim3(BLOCK),
+                         0, stream, meta, found_inf.data_ptr<float>(),
+                         (float)inv_scale);
+    });
+  }
+}

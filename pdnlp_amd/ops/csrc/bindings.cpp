@@ -1,0 +1,71 @@
+// Python bindings for the pdnlp_amd gfx950 kernel extension.
+
+#include <torch/extension.h>
+
+#include <string>
+#include <vector>
+
+std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor w,
+                                         torch::Tensor b, double eps);
+std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
+                                         torch::Tensor w, torch::Tensor mean,
+                                         torch::Tensor rstd);
+std::vector<torch::Tensor> embedding_ln_fwd(
+    torch::Tensor ids, torch::Tensor type_ids, torch::Tensor pos_ids,
+    torch::Tensor word, torch::Tensor pos, torch::Tensor type_,
+    torch::Tensor w, torch::Tensor b, double eps);
+std::vector<torch::Tensor> embedding_ln_bwd(
+    torch::Tensor dy, torch::Tensor ids, torch::Tensor type_ids,
+    torch::Tensor pos_ids, torch::Tensor word, torch::Tensor pos,
+    torch::Tensor type_, torch::Tensor w, torch::Tensor mean,
+    torch::Tensor rstd);
+torch::Tensor masked_softmax_fwd(torch::Tensor scores, torch::Tensor mask,
+                                 double scale);
+torch::Tensor masked_softmax_bwd(torch::Tensor dy, torch::Tensor p);
+torch::Tensor bias_gelu_fwd(torch::Tensor x, torch::Tensor b);
+std::vector<torch::Tensor> bias_gelu_bwd(torch::Tensor dy, torch::Tensor x,
+                                         torch::Tensor b);
+torch::Tensor gelu_bwd(torch::Tensor dy, torch::Tensor pre);
+torch::Tensor tanh_bwd(torch::Tensor dy, torch::Tensor pre);
+std::vector<torch::Tensor> bias_dropout_residual_ln_fwd(
+    torch::Tensor y, torch::Tensor bias, torch::Tensor res, torch::Tensor lnw,
+    torch::Tensor lnb, double p, double eps, long seed);
+std::vector<torch::Tensor> bias_dropout_residual_ln_bwd(
+    torch::Tensor dout, torch::Tensor xsum, torch::Tensor mask,
+    torch::Tensor lnw, torch::Tensor mean, torch::Tensor rstd, double p);
+std::vector<torch::Tensor> cross_entropy_fwd(torch::Tensor logits,
+                                             torch::Tensor labels);
+torch::Tensor cross_entropy_bwd(torch::Tensor dloss, torch::Tensor logprobs,
+                                torch::Tensor labels);
+void multi_tensor_adamw(std::vector<torch::Tensor> params,
+                        std::vector<torch::Tensor> grads,
+                        std::vector<torch::Tensor> exp_avgs,
+                        std::vector<torch::Tensor> exp_avg_sqs,
+                        std::vector<torch::Tensor> masters, double lr,
+                        double beta1, double beta2, double eps,
+                        double weight_decay, double bc1, double bc2,
+                        double grad_scale_inv);
+void multi_tensor_unscale(std::vector<torch::Tensor> grads,
+                          torch::Tensor found_inf, double inv_scale);
+std::vector<torch::Tensor> gemm_nt_fwd(torch::Tensor A, torch::Tensor W,
+                                       torch::Tensor bias, std::string act);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("layernorm_fwd", &layernorm_fwd);
+  m.def("layernorm_bwd", &layernorm_bwd);
+  m.def("embedding_ln_fwd", &embedding_ln_fwd);
+  m.def("embedding_ln_bwd", &embedding_ln_bwd);
+  m.def("masked_softmax_fwd", &masked_softmax_fwd);
+  m.def("masked_softmax_bwd", &masked_softmax_bwd);
+  m.def("bias_gelu_fwd", &bias_gelu_fwd);
+  m.def("bias_gelu_bwd", &bias_gelu_bwd);
+  m.def("gelu_bwd", &gelu_bwd);
+  m.def("tanh_bwd", &tanh_bwd);
+  m.def("bias_dropout_residual_ln_fwd", &bias_dropout_residual_ln_fwd);
+  m.def("bias_dropout_residual_ln_bwd", &bias_dropout_residual_ln_bwd);
+  m.def("cross_entropy_fwd", &cross_entropy_fwd);
+  m.def("cross_entropy_bwd", &cross_entropy_bwd);
+  m.def("multi_tensor_adamw", &multi_tensor_adamw);
+  m.def("multi_tensor_unscale", &multi_tensor_unscale);
+  m.def("gemm_nt_fwd", &gemm_nt_fwd);
+}

@@ -1,0 +1,98 @@
+// Common device helpers for the pdnlp_amd gfx950 kernels.
+//
+// CDNA4 ground rules baked in here (see /opt/skills/guides):
+//  - wavefront = 64 lanes (never 32); block sizes are multiples of 64
+//  - bf16/fp16 global access vectorized as short4/short8 reinterprets
+//  - reductions: wave-level __shfl_xor over 64 lanes, fp32 accumulation
+#pragma once
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <hip/hip_fp16.h>
+
+#define WAVE 64
+
+#define HIP_CHECK(cmd)                                                         \
+  do {                                                                         \
+    hipError_t e = (cmd);                                                      \
+    if (e != hipSuccess) {                                                     \
+      TORCH_CHECK(false, "HIP error: ", hipGetErrorString(e));                 \
+    }                                                                          \
+  } while (0)
+
+// ---- scalar conversions ---------------------------------------------------
+template <typename T> __device__ __forceinline__ float to_f32(T v);
+template <> __device__ __forceinline__ float to_f32<float>(float v) { return v; }
+template <> __device__ __forceinline__ float to_f32<__hip_bfloat16>(__hip_bfloat16 v) {
+  return __bfloat162float(v);
+}
+template <> __device__ __forceinline__ float to_f32<__half>(__half v) {
+  return __half2float(v);
+}
+
+template <typename T> __device__ __forceinline__ T from_f32(float v);
+template <> __device__ __forceinline__ float from_f32<float>(float v) { return v; }
+template <> __device__ __forceinline__ __hip_bfloat16 from_f32<__hip_bfloat16>(float v) {
+  return __float2bfloat16(v);
+}
+template <> __device__ __forceinline__ __half from_f32<__half>(float v) {
+  return __float2half(v);
+}
+
+// ---- wave reductions (64-wide) --------------------------------------------
+__device__ __forceinline__ float wave_sum(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off, WAVE);
+  return v;
+}
+
+__device__ __forceinline__ float wave_max(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v = fmaxf(v, __shfl_xor(v, off, WAVE));
+  return v;
+}
+
+// block reduction over NW waves (NW <= 16), via LDS; returns on all threads
+template <int NW>
+__device__ __forceinline__ float block_sum(float v, float* lds_scratch) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  v = wave_sum(v);
+  if (lane == 0) lds_scratch[wid] = v;
+  __syncthreads();
+  float r = 0.f;
+#pragma unroll
+  for (int i = 0; i < NW; ++i) r += lds_scratch[i];
+  __syncthreads();
+  return r;
+}
+
+// ---- dtype dispatch --------------------------------------------------------
+#define DISPATCH_FLOAT_TYPES(TYPE, NAME, ...)                                  \
+  [&] {                                                                        \
+    switch (TYPE) {                                                            \
+      case at::ScalarType::Float: {                                            \
+        using scalar_t = float;                                                \
+        return __VA_ARGS__();                                                  \
+      }                                                                        \
+      case at::ScalarType::BFloat16: {                                         \
+        using scalar_t = __hip_bfloat16;                                       \
+        return __VA_ARGS__();                                                  \
+      }                                                                        \
+      case at::ScalarType::Half: {                                             \
+        using scalar_t = __half;                                               \
+        return __VA_ARGS__();                                                  \
+      }                                                                        \
+      default:                                                                 \
+        TORCH_CHECK(false, #NAME ": unsupported dtype");                       \
+    }                                                                          \
+  }()
+
+// counter-based RNG for dropout masks (deterministic in (seed, index)).
+__device__ __forceinline__ unsigned int hash_rng(unsigned long long seed,
+                                                 unsigned long long idx) {
+  unsigned long long z = seed * 0x9E3779B97F4A7C15ull + idx + 1ull;
+  z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ull;
+  z = (z ^ (z >> 27)) * 0x94D049BB133111EBull;
+  return (unsigned int)(z >> 33);
+}

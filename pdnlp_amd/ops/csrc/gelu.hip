@@ -1,0 +1,159 @@
+// Bias+GELU forward and GELU/tanh backward epilogue kernels.
+// Exact erf GELU (HF "gelu"). Grid-stride, 8 elements per thread iteration,
+// vectorized 16-byte loads for bf16/fp16 (G13: hipcc does not auto-vectorize
+// scalar bf16 loads).
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "common.h"
+
+namespace {
+
+__device__ __forceinline__ float gelu_f(float x) {
+  return 0.5f * x * (1.f + erff(x * 0.70710678118654752f));
+}
+
+__device__ __forceinline__ float gelu_grad_f(float x) {
+  const float cdf = 0.5f * (1.f + erff(x * 0.70710678118654752f));
+  const float pdf = 0.3989422804014327f * __expf(-0.5f * x * x);
+  return cdf + x * pdf;
+}
+
+template <typename T>
+__global__ void bias_gelu_fwd_kernel(const T* __restrict__ x,
+                                     const T* __restrict__ b,
+                                     T* __restrict__ y, long total, int N) {
+  const long i0 = (long)(blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  const long stride = (long)gridDim.x * blockDim.x * 8;
+  for (long i = i0; i < total; i += stride) {
+    if (sizeof(T) == 2 && i + 8 <= total) {
+      short4 r0 = *reinterpret_cast<const short4*>(x + i);
+      short4 r1 = *reinterpret_cast<const short4*>(x + i + 4);
+      const T* px0 = reinterpret_cast<const T*>(&r0);
+      const T* px1 = reinterpret_cast<const T*>(&r1);
+      short4 o0, o1;
+      T* py0 = reinterpret_cast<T*>(&o0);
+      T* py1 = reinterpret_cast<T*>(&o1);
+      const int c = (int)(i % N);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        py0[j] = from_f32<T>(
+            gelu_f(to_f32<T>(px0[j]) + to_f32<T>(b[(c + j) % N])));
+        py1[j] = from_f32<T>(
+            gelu_f(to_f32<T>(px1[j]) + to_f32<T>(b[(c + 4 + j) % N])));
+      }
+      *reinterpret_cast<short4*>(y + i) = o0;
+      *reinterpret_cast<short4*>(y + i + 4) = o1;
+    } else {
+      for (long k = i; k < total; ++k) {
+        y[k] = from_f32<T>(gelu_f(to_f32<T>(x[k]) + to_f32<T>(b[k % N])));
+      }
+    }
+  }
+}
+
+template <typename T>
+__global__ void bias_gelu_bwd_kernel(const T* __restrict__ dy,
+                                     const T* __restrict__ x,
+                                     const T* __restrict__ b,
+                                     T* __restrict__ dx, long total, int N) {
+  const long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i = i0; i < total; i += stride) {
+    const float pre = to_f32<T>(x[i]) + to_f32<T>(b[i % N]);
+    dx[i] = from_f32<T>(to_f32<T>(dy[i]) * gelu_grad_f(pre));
+  }
+}
+
+// dgelu on a saved pre-activation (GEMM fused-epilogue backward)
+template <typename T>
+__global__ void gelu_bwd_kernel(const T* __restrict__ dy,
+                                const T* __restrict__ pre,
+                                T* __restrict__ dx, long total) {
+  const long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i = i0; i < total; i += stride) {
+    dx[i] = from_f32<T>(to_f32<T>(dy[i]) * gelu_grad_f(to_f32<T>(pre[i])));
+  }
+}
+
+template <typename T>
+__global__ void tanh_bwd_kernel(const T* __restrict__ dy,
+                                const T* __restrict__ pre,
+                                T* __restrict__ dx, long total) {
+  const long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i = i0; i < total; i += stride) {
+    const float t = tanhf(to_f32<T>(pre[i]));
+    dx[i] = from_f32<T>(to_f32<T>(dy[i]) * (1.f - t * t));
+  }
+}
+
+int grid_for(long total, int per_thread = 1) {
+  const long want = (total + 256L * per_thread - 1) / (256L * per_thread);
+  return (int)std::min<long>(want, 2048);  // G11: cap + grid-stride
+}
+
+}  // namespace
+
+torch::Tensor bias_gelu_fwd(torch::Tensor x, torch::Tensor b) {
+  auto y = torch::empty_like(x);
+  const long total = x.numel();
+  const int N = b.numel();
+  auto stream = at::hip::getCurrentHIPStream();
+  DISPATCH_FLOAT_TYPES(x.scalar_type(), "bias_gelu_fwd", [&] {
+    hipLaunchKernelGGL((bias_gelu_fwd_kernel<scalar_t>),
+                       dim3(grid_for(total, 8)), dim3(256), 0, stream,
+                       (const scalar_t*)x.data_ptr(),
+                       (const scalar_t*)b.data_ptr(), (scalar_t*)y.data_ptr(),
+                       total, N);
+  });
+  return y;
+}
+
+std::vector<torch::Tensor> bias_gelu_bwd(torch::Tensor dy, torch::Tensor x,
+                                         torch::Tensor b) {
+  auto dx = torch::empty_like(x);
+  const long total = x.numel();
+  const int N = b.numel();
+  auto stream = at::hip::getCurrentHIPStream();
+  DISPATCH_FLOAT_TYPES(x.scalar_type(), "bias_gelu_bwd", [&] {
+    hipLaunchKernelGGL((bias_gelu_bwd_kernel<scalar_t>),
+                       dim3(grid_for(total)), dim3(256), 0, stream,
+                       (const scalar_t*)dy.data_ptr(),
+                       (const scalar_t*)x.data_ptr(),
+                       (const scalar_t*)b.data_ptr(),
+                       (scalar_t*)dx.data_ptr(), total, N);
+  });
+  auto db = dx.view({-1, N}).sum(0);
+  return {dx, db};
+}
+
+torch::Tensor gelu_bwd(torch::Tensor dy, torch::Tensor pre) {
+  auto dx = torch::empty_like(dy);
+  const long total = dy.numel();
+  auto stream = at::hip::getCurrentHIPStream();
+  DISPATCH_FLOAT_TYPES(dy.scalar_type(), "gelu_bwd", [&] {
+    hipLaunchKernelGGL((gelu_bwd_kernel<scalar_t>), dim3(grid_for(total)),
+                       dim3(256), 0, stream,
+                       (const scalar_t*)dy.data_ptr(),
+                       (const scalar_t*)pre.data_ptr(),
+                       (scalar_t*)dx.data_ptr(), total);
+  });
+  return dx;
+}
+
+torch::Tensor tanh_bwd(torch::Tensor dy, torch::Tensor pre) {
+  auto dx = torch::empty_like(dy);
+  const long total = dy.numel();
+  auto stream = at::hip::getCurrentHIPStream();
+  DISPATCH_FLOAT_TYPES(dy.scalar_type(), "tanh_bwd", [&] {
+    hipLaunchKernelGGL((tanh_bwd_kernel<scalar_t>), dim3(grid_for(total)),
+                       dim3(256), 0, stream,
+                       (const scalar_t*)dy.data_ptr(),
+                       (const scalar_t*)pre.data_ptr(),
+                       (scalar_t*)dx.data_ptr(), total);
+  });
+  return dx;
+}

@@ -1,0 +1,230 @@
+// Hand-written CDNA4 MFMA GEMM: C[M,N] = A[M,K] @ W[N,K]^T (+bias) (+act)
+// — the forward GEMMs of the BERT hot path with
+// the bias/GELU/tanh epilogue fused into the C-write.
+//
+// Structure (the "step-3" ladder structure of the CDNA4 guide, §5):
+//  - 128x128 output tile, BK=64, 4 waves (256 threads), each wave a 64x64
+//    sub-tile = 4x4 fragments of v_mfma_f32_16x16x32_bf16 (f16 variant for
+//    fp16), fp32 accumulation in AGPRs.
+//  - global->LDS staging via __builtin_amdgcn_global_load_lds width 16
+//    (2 LDS buffers, stage tile t+1 while computing tile t, one
+//    vmcnt(0)+barrier per tile).
+//  - LDS bank-conflict fix: XOR swizzle byte ^= ((row&7)<<4) applied on the
+//    *global source address* (glds writes lane-linear; guide §5.4 rule 21)
+//    and on the ds_read fragment address — ≤2-way instead of 8-way.
+//  - XCD-aware block swizzle (T1): contiguous tile chunks per XCD for L2
+//    affinity (bijective variant).
+//
+// A-fragment: lane l holds A[l&15][(l>>4)*8 + i], i=0..7 -> one ds_read_b128.
+// B-fragment: lane l holds W[n0 + (l&15)][k0 + (l>>4)*8 + i] — the same
+// pattern because W is stored [N,K] row-major (torch Linear layout) and
+// C = A @ W^T. C/D: col = lane&15, row = (lane>>4)*4 + reg.
+//
+// Backward dGEMMs (dX = dY@W, dW = dY^T@X) are plain non-fused GEMMs and go
+// through rocBLAS (torch.matmul) from functional.py.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "common.h"
+
+namespace {
+
+typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
+typedef _Float16 f16x8 __attribute__((ext_vector_type(8)));
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+
+constexpr int BM = 128, BN = 128, BK = 64;
+constexpr int NTHREADS = 256;
+
+__device__ __forceinline__ float gelu_f2(float x) {
+  return 0.5f * x * (1.f + erff(x * 0.70710678118654752f));
+}
+
+// stage a BM x BK tile (rows of `src`, row stride `ld` elements) into LDS via
+// global_load_lds. Linear LDS image [128 rows][128 bytes]; the XOR swizzle is
+// pre-applied on the source byte offset. Each wave-instruction moves 8 rows
+// (8 lanes of 16 B per row); 4 waves x 4 calls cover 128 rows.
+template <typename T>
+__device__ __forceinline__ void stage_tile(const T* __restrict__ src, long ld,
+                                           long row0, long max_row, long k0,
+                                           char* lds) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x >> 6;
+  const int sub_row = lane >> 3;           // 0..7
+  const int piece = lane & 7;              // 16B piece within the 128B row
+  const int kbyte = (piece * 16) ^ (sub_row << 4);  // source pre-swizzle
+#pragma unroll
+  for (int c = 0; c < 4; ++c) {
+    const int r = (wid * 4 + c) * 8 + sub_row;   // tile row 0..127
+    long gr = row0 + r;
+    gr = gr < max_row ? gr : max_row - 1;        // clamp tail (stores guard)
+    const char* gp = (const char*)(src + gr * ld + k0) + kbyte;
+    // LDS dest operand is WAVE-UNIFORM (start of this call's 8-row chunk);
+    // hardware writes lane l at dest + l*16 = row (l>>3), piece (l&7).
+    char* lp = lds + (long)(wid * 4 + c) * 8 * 128;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)gp,
+        (__attribute__((address_space(3))) unsigned int*)lp, 16, 0, 0);
+  }
+}
+
+// read one 16x(8k) fragment from the swizzled LDS image
+template <typename V8>
+__device__ __forceinline__ V8 read_frag(const char* lds, int frag_row0,
+                                        int ks) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int row = frag_row0 + (lane & 15);
+  const int colbyte = (ks * 64 + (lane >> 4) * 16) ^ ((row & 7) << 4);
+  return *reinterpret_cast<const V8*>(lds + row * 128 + colbyte);
+}
+
+enum Act { ACT_NONE = 0, ACT_GELU = 1, ACT_TANH = 2 };
+
+template <typename T, typename V8, bool HAS_BIAS, int ACT, bool SAVE_PRE>
+__global__ __launch_bounds__(NTHREADS)
+void gemm_nt_kernel(const T* __restrict__ A, const T* __restrict__ W,
+                    const T* __restrict__ bias, T* __restrict__ C,
+                    T* __restrict__ pre, long M, long N, long K,
+                    int tiles_n, int nwg) {
+  // XCD-aware bijective remap of the tile id (guide T1)
+  int wg = blockIdx.x;
+  {
+    const int nxcd = 8;
+    const int q = nwg / nxcd, r = nwg % nxcd;
+    const int xcd = wg % nxcd, idx = wg / nxcd;
+    wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  }
+  const long tile_m = wg / tiles_n, tile_n = wg % tiles_n;
+  const long m0 = tile_m * BM, n0 = tile_n * BN;
+
+  __shared__ __attribute__((aligned(16))) char lds[2][2][BM * 128];  // [buf][A/B]
+
+  const int wid = threadIdx.x >> 6;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wr = (wid >> 1) * 64, wc = (wid & 1) * 64;  // wave sub-tile origin
+
+  f32x4 acc[4][4] = {};
+
+  stage_tile<T>(A, K, m0, M, 0, lds[0][0]);
+  stage_tile<T>(W, K, n0, N, 0, lds[0][1]);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+
+  const int ntiles = (int)(K / BK);
+  int cur = 0;
+  for (int t = 0; t < ntiles; ++t) {
+    if (t + 1 < ntiles) {
+      stage_tile<T>(A, K, m0, M, (long)(t + 1) * BK, lds[cur ^ 1][0]);
+      stage_tile<T>(W, K, n0, N, (long)(t + 1) * BK, lds[cur ^ 1][1]);
+    }
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      V8 a_frag[4], b_frag[4];
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+        a_frag[i] = read_frag<V8>(lds[cur][0], wr + i * 16, ks);
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        b_frag[j] = read_frag<V8>(lds[cur][1], wc + j * 16, ks);
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          if constexpr (std::is_same<V8, bf16x8>::value) {
+            acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
+          } else {
+            acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_f16(
+                a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
+          }
+        }
+      }
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  // epilogue: bias + activation fused into the C write
+  const int crow_off = (lane >> 4) * 4;
+  const int ccol = lane & 15;
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const long n = n0 + wc + j * 16 + ccol;
+      if (n >= N) continue;
+      const float bv = HAS_BIAS ? to_f32<T>(bias[n]) : 0.f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const long m = m0 + wr + i * 16 + crow_off + r;
+        if (m >= M) continue;
+        float v = acc[i][j][r] + bv;
+        if (SAVE_PRE) pre[m * N + n] = from_f32<T>(v);
+        if (ACT == ACT_GELU) v = gelu_f2(v);
+        if (ACT == ACT_TANH) v = tanhf(v);
+        C[m * N + n] = from_f32<T>(v);
+      }
+    }
+  }
+}
+
+template <typename T, typename V8>
+void launch_gemm(const torch::Tensor& A, const torch::Tensor& W,
+                 const torch::Tensor& bias, torch::Tensor& C,
+                 torch::Tensor& pre, int act, bool has_bias, bool save_pre,
+                 hipStream_t stream) {
+  const long M = A.size(0), K = A.size(1), N = W.size(0);
+  const int tiles_m = (int)((M + BM - 1) / BM);
+  const int tiles_n = (int)((N + BN - 1) / BN);
+  const int nwg = tiles_m * tiles_n;
+  const T* bptr = has_bias ? (const T*)bias.data_ptr() : nullptr;
+  T* pptr = save_pre ? (T*)pre.data_ptr() : nullptr;
+
+#define LAUNCH(HB, ACTV, SP)                                                   \
+  hipLaunchKernelGGL((gemm_nt_kernel<T, V8, HB, ACTV, SP>), dim3(nwg),         \
+                     dim3(NTHREADS), 0, stream, (const T*)A.data_ptr(),        \
+                     (const T*)W.data_ptr(), bptr, (T*)C.data_ptr(), pptr, M,  \
+                     N, K, tiles_n, nwg)
+
+  if (act == ACT_NONE) {
+    if (has_bias) LAUNCH(true, ACT_NONE, false);
+    else LAUNCH(false, ACT_NONE, false);
+  } else if (act == ACT_GELU) {
+    if (has_bias) LAUNCH(true, ACT_GELU, true);
+    else LAUNCH(false, ACT_GELU, true);
+  } else {
+    if (has_bias) LAUNCH(true, ACT_TANH, true);
+    else LAUNCH(false, ACT_TANH, true);
+  }
+#undef LAUNCH
+}
+
+}  // namespace
+
+std::vector<torch::Tensor> gemm_nt_fwd(torch::Tensor A, torch::Tensor W,
+                                       torch::Tensor bias, std::string act) {
+  TORCH_CHECK(A.is_cuda() && A.is_contiguous() && W.is_contiguous());
+  TORCH_CHECK(A.dim() == 2 && W.dim() == 2 && A.size(1) == W.size(1));
+  TORCH_CHECK(A.size(1) % BK == 0, "K must be a multiple of 64");
+  const long M = A.size(0), N = W.size(0);
+  const bool has_bias = bias.defined() && bias.numel() > 0;
+  const int actv = act == "gelu" ? ACT_GELU : act == "tanh" ? ACT_TANH
+                                                            : ACT_NONE;
+  const bool save_pre = actv != ACT_NONE;
+  auto C = torch::empty({M, N}, A.options());
+  auto pre = save_pre ? torch::empty({M, N}, A.options())
+                      : torch::empty({0}, A.options());
+  auto stream = at::hip::getCurrentHIPStream();
+  if (A.scalar_type() == torch::kBFloat16) {
+    launch_gemm<__hip_bfloat16, bf16x8>(A, W, bias, C, pre, actv, has_bias,
+                                        save_pre, stream);
+  } else if (A.scalar_type() == torch::kHalf) {
+    launch_gemm<__half, f16x8>(A, W, bias, C, pre, actv, has_bias, save_pre,
+                               stream);
+  } else {
+    TORCH_CHECK(false, "gemm_nt_fwd: bf16/fp16 only");
+  }
+  return {C, pre};
+}

@@ -18,8 +18,11 @@ from torch.utils import cpp_extension  # noqa: E402
 
 CSRC = os.path.join(os.path.dirname(os.path.abspath(__file__)),
                     "pdnlp_amd", "ops", "csrc")
+# hipify writes X_hip.hip copies next to X.hip — exclude them from the glob
+# (they are build artifacts, also git-ignored)
 sources = sorted(glob.glob(os.path.join(CSRC, "*.cpp"))
-                 + glob.glob(os.path.join(CSRC, "*.hip")))
+                 + [f for f in glob.glob(os.path.join(CSRC, "*.hip"))
+                    if not f.endswith("_hip.hip")])
 
 ext_modules = []
 if sources:

@@ -1,0 +1,308 @@
+"""GPU numerics tests: every HIP kernel vs a plain PyTorch fp32 reference
+(the torch-path composition in functional.py). fp32 kernel runs get tight
+tolerances; bf16 runs get bf16-roundoff tolerances."""
+
+import math
+
+import pytest
+import torch
+import torch.nn.functional as F
+
+pytestmark = [pytest.mark.gpu, pytest.mark.timeout(600)]
+
+
+def ext():
+    from pdnlp_amd.ops import ext as _ext
+    e = _ext()
+    assert e is not None, "HIP extension must be built on the GPU box"
+    return e
+
+
+DEV = "cuda:0"
+
+
+# ---------------------------------------------------------------- layernorm
+@pytest.mark.parametrize("dtype,rtol,atol", [
+    (torch.float32, 1e-5, 1e-5), (torch.bfloat16, 2e-2, 2e-2)])
+@pytest.mark.parametrize("H", [64, 768, 1024])
+def test_layernorm_fwd_bwd(dtype, rtol, atol, H):
+    e = ext()
+    torch.manual_seed(0)
+    R = 512
+    x = torch.randn(R, H, device=DEV, dtype=dtype)
+    w = torch.randn(H, device=DEV, dtype=dtype)
+    b = torch.randn(H, device=DEV, dtype=dtype)
+    y, mean, rstd = e.layernorm_fwd(x, w, b, 1e-12)
+    ref = F.layer_norm(x.float(), (H,), w.float(), b.float(), 1e-12)
+    torch.testing.assert_close(y.float(), ref, rtol=rtol, atol=atol)
+
+    dy = torch.randn_like(x)
+    dx, dw, db = e.layernorm_bwd(dy, x, w, mean, rstd)
+    xr = x.float().detach().requires_grad_()
+    wr = w.float().detach().requires_grad_()
+    br = b.float().detach().requires_grad_()
+    F.layer_norm(xr, (H,), wr, br, 1e-12).backward(dy.float())
+    torch.testing.assert_close(dx.float(), xr.grad, rtol=rtol, atol=atol * 10)
+    torch.testing.assert_close(dw.float(), wr.grad, rtol=rtol, atol=atol * 50)
+    torch.testing.assert_close(db.float(), br.grad, rtol=rtol, atol=atol * 50)
+
+
+# ------------------------------------------------------------- embedding+LN
+@pytest.mark.parametrize("dtype,rtol,atol", [
+    (torch.float32, 1e-5, 1e-5), (torch.bfloat16, 2e-2, 2e-2)])
+def test_embedding_ln(dtype, rtol, atol):
+    e = ext()
+    torch.manual_seed(1)
+    B, S, H, V = 4, 32, 768, 1000
+    ids = torch.randint(0, V, (B, S), device=DEV)
+    tids = torch.randint(0, 2, (B, S), device=DEV)
+    pids = torch.arange(S, device=DEV).repeat(B, 1).contiguous()
+    word = torch.randn(V, H, device=DEV, dtype=dtype)
+    pos = torch.randn(S, H, device=DEV, dtype=dtype)
+    typ = torch.randn(2, H, device=DEV, dtype=dtype)
+    lnw = torch.randn(H, device=DEV, dtype=dtype)
+    lnb = torch.randn(H, device=DEV, dtype=dtype)
+    y, mean, rstd = e.embedding_ln_fwd(ids, tids, pids, word, pos, typ,
+                                       lnw, lnb, 1e-12)
+    emb = (F.embedding(ids, word.float()) + F.embedding(pids, pos.float())
+           + F.embedding(tids, typ.float()))
+    ref = F.layer_norm(emb, (H,), lnw.float(), lnb.float(), 1e-12)
+    torch.testing.assert_close(y.float(), ref, rtol=rtol, atol=atol)
+
+    dy = torch.randn(B, S, H, device=DEV, dtype=dtype)
+    dword, dpos, dtyp, dlnw, dlnb = e.embedding_ln_bwd(
+        dy, ids, tids, pids, word, pos, typ, lnw, mean, rstd)
+    wr = word.float().detach().requires_grad_()
+    pr = pos.float().detach().requires_grad_()
+    tr = typ.float().detach().requires_grad_()
+    lw = lnw.float().detach().requires_grad_()
+    lb = lnb.float().detach().requires_grad_()
+    emb = F.embedding(ids, wr) + F.embedding(pids, pr) + F.embedding(tids, tr)
+    F.layer_norm(emb, (H,), lw, lb, 1e-12).backward(dy.float())
+    torch.testing.assert_close(dword.float(), wr.grad, rtol=rtol, atol=atol * 20)
+    torch.testing.assert_close(dpos.float(), pr.grad, rtol=rtol, atol=atol * 20)
+    torch.testing.assert_close(dtyp.float(), tr.grad, rtol=rtol, atol=atol * 50)
+    torch.testing.assert_close(dlnw.float(), lw.grad, rtol=rtol, atol=atol * 50)
+    torch.testing.assert_close(dlnb.float(), lb.grad, rtol=rtol, atol=atol * 50)
+
+
+# ------------------------------------------------------------ masked softmax
+@pytest.mark.parametrize("dtype,rtol,atol", [
+    (torch.float32, 1e-5, 1e-6), (torch.bfloat16, 1e-2, 1e-2)])
+@pytest.mark.parametrize("S", [128, 512])
+def test_masked_softmax(dtype, rtol, atol, S):
+    e = ext()
+    torch.manual_seed(2)
+    B, NH = 4, 12
+    scores = torch.randn(B, NH, S, S, device=DEV, dtype=dtype) * 4
+    mask = torch.zeros(B, 1, 1, S, device=DEV, dtype=dtype)
+    mask[:, :, :, S // 2:] = -10000.0
+    scale = 1.0 / math.sqrt(64)
+    p = e.masked_softmax_fwd(scores, mask, scale)
+    ref = F.softmax(scores.float() * scale + mask.float(), dim=-1)
+    torch.testing.assert_close(p.float(), ref, rtol=rtol, atol=atol)
+    assert torch.allclose(p.float().sum(-1),
+                          torch.ones(B, NH, S, S // S, device=DEV).squeeze(-1),
+                          atol=1e-2)
+
+    dy = torch.randn_like(scores)
+    dx = e.masked_softmax_bwd(dy, p)
+    sref = (scores.float() * scale + mask.float()).detach().requires_grad_()
+    F.softmax(sref, dim=-1).backward(dy.float())
+    torch.testing.assert_close(dx.float(), sref.grad, rtol=rtol, atol=atol)
+
+
+# ------------------------------------------------------------------ gelu ops
+@pytest.mark.parametrize("dtype,rtol,atol", [
+    (torch.float32, 1e-5, 1e-6), (torch.bfloat16, 2e-2, 2e-2)])
+def test_bias_gelu(dtype, rtol, atol):
+    e = ext()
+    torch.manual_seed(3)
+    R, N = 512, 3072
+    x = torch.randn(R, N, device=DEV, dtype=dtype)
+    b = torch.randn(N, device=DEV, dtype=dtype)
+    y = e.bias_gelu_fwd(x, b)
+    ref = F.gelu(x.float() + b.float())
+    torch.testing.assert_close(y.float(), ref, rtol=rtol, atol=atol)
+
+    dy = torch.randn_like(x)
+    dx, db = e.bias_gelu_bwd(dy, x, b)
+    xr = x.float().detach().requires_grad_()
+    br = b.float().detach().requires_grad_()
+    F.gelu(xr + br).backward(dy.float())
+    torch.testing.assert_close(dx.float(), xr.grad, rtol=rtol, atol=atol)
+    torch.testing.assert_close(db.float(), br.grad, rtol=rtol, atol=atol * 100)
+
+
+# ------------------------------------------- bias+dropout+residual+layernorm
+@pytest.mark.parametrize("dtype,rtol,atol", [
+    (torch.float32, 1e-5, 1e-5), (torch.bfloat16, 2e-2, 2e-2)])
+def test_bdrl_no_dropout(dtype, rtol, atol):
+    e = ext()
+    torch.manual_seed(4)
+    R, H = 512, 768
+    y = torch.randn(R, H, device=DEV, dtype=dtype)
+    bias = torch.randn(H, device=DEV, dtype=dtype)
+    res = torch.randn(R, H, device=DEV, dtype=dtype)
+    lnw = torch.randn(H, device=DEV, dtype=dtype)
+    lnb = torch.randn(H, device=DEV, dtype=dtype)
+    out, xsum, mask, mean, rstd = e.bias_dropout_residual_ln_fwd(
+        y, bias, res, lnw, lnb, 0.0, 1e-12, 0)
+    ref = F.layer_norm(y.float() + bias.float() + res.float(), (H,),
+                       lnw.float(), lnb.float(), 1e-12)
+    torch.testing.assert_close(out.float(), ref, rtol=rtol, atol=atol)
+
+    dout = torch.randn_like(y)
+    dy, dbias, dres, dlnw, dlnb = e.bias_dropout_residual_ln_bwd(
+        dout, xsum, mask, lnw, mean, rstd, 0.0)
+    yr = y.float().detach().requires_grad_()
+    br = bias.float().detach().requires_grad_()
+    rr = res.float().detach().requires_grad_()
+    lw = lnw.float().detach().requires_grad_()
+    lb = lnb.float().detach().requires_grad_()
+    F.layer_norm(yr + br + rr, (H,), lw, lb, 1e-12).backward(dout.float())
+    torch.testing.assert_close(dy.float(), yr.grad, rtol=rtol, atol=atol * 10)
+    torch.testing.assert_close(dres.float(), rr.grad, rtol=rtol, atol=atol * 10)
+    torch.testing.assert_close(dbias.float(), br.grad, rtol=5e-2, atol=atol * 100)
+    torch.testing.assert_close(dlnw.float(), lw.grad, rtol=rtol, atol=atol * 50)
+    torch.testing.assert_close(dlnb.float(), lb.grad, rtol=rtol, atol=atol * 50)
+
+
+def test_bdrl_dropout_statistics():
+    e = ext()
+    torch.manual_seed(5)
+    R, H, p = 2048, 768, 0.1
+    y = torch.randn(R, H, device=DEV, dtype=torch.float32)
+    z = torch.zeros(H, device=DEV)
+    res = torch.zeros(R, H, device=DEV)
+    lnw = torch.ones(H, device=DEV)
+    lnb = torch.zeros(H, device=DEV)
+    out, xsum, mask, mean, rstd = e.bias_dropout_residual_ln_fwd(
+        y, z, res, lnw, lnb, p, 1e-12, 1234)
+    keep_rate = mask.float().mean().item()
+    assert abs(keep_rate - (1 - p)) < 5e-3, keep_rate
+    # kept elements are scaled by 1/(1-p) before the residual add
+    kept = mask.view(R, H).bool()
+    torch.testing.assert_close(xsum[kept], y[kept] / (1 - p),
+                               rtol=1e-5, atol=1e-5)
+    assert (xsum[~kept] == 0).all()
+    # determinism in seed
+    out2, xsum2, mask2, _, _ = e.bias_dropout_residual_ln_fwd(
+        y, z, res, lnw, lnb, p, 1e-12, 1234)
+    assert torch.equal(mask, mask2)
+
+
+# ---------------------------------------------------------------------- gemm
+@pytest.mark.parametrize("M,N,K", [(256, 768, 768), (4096, 3072, 768),
+                                   (4096, 768, 3072), (32, 768, 768),
+                                   (100, 128, 64)])
+def test_gemm_nt_bf16(M, N, K):
+    e = ext()
+    torch.manual_seed(6)
+    A = (torch.randn(M, K, device=DEV) / math.sqrt(K)).bfloat16()
+    W = torch.randn(N, K, device=DEV).bfloat16()
+    bias = torch.randn(N, device=DEV).bfloat16()
+    C, _ = e.gemm_nt_fwd(A, W, bias, "none")
+    ref = A.float() @ W.float().t() + bias.float()
+    err = (C.float() - ref).abs().max().item()
+    scale = ref.abs().max().item()
+    assert err < 2e-2 * max(scale, 1.0), (err, scale)
+
+
+def test_gemm_identity_and_transpose_detection():
+    """Asymmetric-B identity test (guide: symmetric inputs miss transposes)."""
+    e = ext()
+    K = 64
+    A = torch.eye(128, K, device=DEV).bfloat16()
+    W = torch.zeros(128, K, device=DEV)
+    for i in range(128):
+        for j in range(0, K, 7):
+            W[i, j] = i * 0.01 + j  # asymmetric
+    W = W.bfloat16()
+    C, _ = e.gemm_nt_fwd(A, W, torch.Tensor().to(DEV), "none")
+    ref = A.float() @ W.float().t()
+    torch.testing.assert_close(C.float(), ref, rtol=1e-3, atol=1e-3)
+
+
+def test_gemm_gelu_epilogue():
+    e = ext()
+    torch.manual_seed(7)
+    M, N, K = 512, 3072, 768
+    A = (torch.randn(M, K, device=DEV) / math.sqrt(K)).bfloat16()
+    W = torch.randn(N, K, device=DEV).bfloat16()
+    bias = torch.randn(N, device=DEV).bfloat16()
+    C, pre = e.gemm_nt_fwd(A, W, bias, "gelu")
+    pre_ref = A.float() @ W.float().t() + bias.float()
+    ref = F.gelu(pre_ref)
+    assert (pre.float() - pre_ref).abs().max().item() < 0.05
+    assert (C.float() - ref).abs().max().item() < 0.05
+
+
+# --------------------------------------------------------------------- adamw
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_multi_tensor_adamw(dtype):
+    import pdnlp_amd.ops.adamw as A
+    torch.manual_seed(8)
+    shapes = [(768,), (768, 768), (3072,), (21128, 768), (5,)]
+    params = [torch.randn(*s, device=DEV, dtype=dtype) for s in shapes]
+    grads = [torch.randn(*s, device=DEV, dtype=dtype) for s in shapes]
+    ms = [torch.rand(*s, device=DEV) * 0.1 for s in shapes]
+    vs = [torch.rand(*s, device=DEV) * 0.01 for s in shapes]
+    use_master = dtype != torch.float32
+    masters = [p.float().clone() if use_master else None for p in params]
+
+    ref_p = [p.clone() for p in params]
+    ref_m = [m.clone() for m in ms]
+    ref_v = [v.clone() for v in vs]
+    ref_mw = [mw.clone() if mw is not None else None for mw in masters]
+
+    e = ext()
+    e.multi_tensor_adamw(params, grads, ms, vs,
+                         masters if use_master else [],
+                         1e-3, 0.9, 0.999, 1e-8, 0.01, 0.1, 0.001, 1.0)
+    # python reference (the fallback in ops.adamw)
+    import os
+    os.environ["PDNLP_FORCE_TORCH"] = "1"
+    try:
+        A.multi_tensor_adamw(ref_p, grads, ref_m, ref_v, ref_mw,
+                             1e-3, 0.9, 0.999, 1e-8, 0.01,
+                             step=1, grad_scale_inv=1.0)
+    finally:
+        del os.environ["PDNLP_FORCE_TORCH"]
+    # step=1 with betas -> bc1=0.1, bc2=0.001 matches the raw bc args above
+    for p, rp in zip(params, ref_p):
+        tol = 1e-6 if dtype == torch.float32 else 1e-2
+        torch.testing.assert_close(p.float(), rp.float(), rtol=tol, atol=tol)
+    for m, rm in zip(ms, ref_m):
+        torch.testing.assert_close(m, rm, rtol=1e-5, atol=1e-6)
+
+
+def test_multi_tensor_unscale():
+    e = ext()
+    g1 = torch.full((1000,), 8.0, device=DEV)
+    g2 = torch.full((37,), 4.0, device=DEV)
+    found = torch.zeros(1, device=DEV)
+    e.multi_tensor_unscale([g1, g2], found, 0.25)
+    assert found.item() == 0
+    torch.testing.assert_close(g1, torch.full((1000,), 2.0, device=DEV))
+    torch.testing.assert_close(g2, torch.full((37,), 1.0, device=DEV))
+    g1[500] = float("inf")
+    e.multi_tensor_unscale([g1], found, 1.0)
+    assert found.item() == 1
+
+
+# ------------------------------------------------------------- cross entropy
+def test_cross_entropy():
+    e = ext()
+    torch.manual_seed(9)
+    B, C = 32, 6
+    logits = torch.randn(B, C, device=DEV)
+    labels = torch.randint(0, C, (B,), device=DEV)
+    loss, logprobs = e.cross_entropy_fwd(logits, labels)
+    ref = F.cross_entropy(logits, labels)
+    torch.testing.assert_close(loss, ref, rtol=1e-5, atol=1e-6)
+    dloss = torch.tensor(1.7, device=DEV)
+    dl = e.cross_entropy_bwd(dloss, logprobs, labels)
+    lr = logits.detach().requires_grad_()
+    (F.cross_entropy(lr, labels) * 1.7).backward()
+    torch.testing.assert_close(dl, lr.grad, rtol=1e-5, atol=1e-6)
