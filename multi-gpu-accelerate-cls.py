@@ -36,7 +36,7 @@ def main():
         args.rank = dist.get_rank()
         args.world_size = dist.get_world_size()
     train_loader, dev_loader, _ = build_dataloaders(args, 1, 0)  # unsharded
-    model = build_model("bert-base", model_path=args.model_path)
+    model = build_model(args.model, model_path=args.model_path)
     optimizer = build_optimizer(model, lr=args.learning_rate,
                                 weight_decay=args.weight_decay)
     model, optimizer, train_loader, dev_loader = accelerator.prepare(

@@ -50,7 +50,7 @@ def main():
         dev_ds = SyntheticClsDataset(base.data_limit - n_train,
                                      base.max_seq_len, seed=base.seed + 1)
         collate = Collate(None, base.max_seq_len, label_key="labels")
-    model = build_model("bert-base", model_path=base.model_path)
+    model = build_model(base.model, model_path=base.model_path)
     trainer = HFStyleTrainer(model, hf_args, train_dataset=train_ds,
                              eval_dataset=dev_ds, data_collator=collate,
                              compute_metrics=compute_metrics)

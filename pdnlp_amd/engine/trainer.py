@@ -48,7 +48,7 @@ class Trainer:
         self.world = dist.get_world_size() if dist.is_initialized() else 1
         self.metrics = MetricsWriter(getattr(args, "metrics_jsonl", None),
                                      rank=self.rank)
-        self.best_acc = 0.0
+        self.best_acc = -1.0  # first dev eval always checkpoints
         self.global_step = 0
 
     # ------------------------------------------------------------------

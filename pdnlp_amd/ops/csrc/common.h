@@ -94,5 +94,5 @@ __device__ __forceinline__ unsigned int hash_rng(unsigned long long seed,
   unsigned long long z = seed * 0x9E3779B97F4A7C15ull + idx + 1ull;
   z = (z ^ (z >> 30)) * 0xBF58476D1CE4E5B9ull;
   z = (z ^ (z >> 27)) * 0x94D049BB133111EBull;
-  return (unsigned int)(z >> 33);
+  return (unsigned int)(z >> 32);
 }

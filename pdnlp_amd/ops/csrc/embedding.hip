@@ -164,8 +164,8 @@ std::vector<torch::Tensor> embedding_ln_bwd(
   auto dlnw32 = torch::zeros({H}, opts32);
   auto dlnb32 = torch::zeros({H}, opts32);
   auto stream = at::hip::getCurrentHIPStream();
-  const int chunks = std::max<long>(1, std::min<long>(32, R / 1024));
-  const int rows_per_chunk = (R + chunks - 1) / chunks;
+  const int rows_per_chunk = 16;
+  const int chunks = (int)((R + rows_per_chunk - 1) / rows_per_chunk);
   DISPATCH_FLOAT_TYPES(word.scalar_type(), "embedding_ln_bwd", [&] {
     hipLaunchKernelGGL((emb_ln_bwd_scatter_kernel<scalar_t>), dim3(R),
                        dim3(WAVE), 0, stream,

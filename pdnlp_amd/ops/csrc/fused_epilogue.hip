@@ -104,27 +104,34 @@ __global__ void bdrl_bwd_dx_kernel(const T* __restrict__ dout,
   }
 }
 
+// also folds the projection-bias gradient (column sum of dy, the
+// post-dropout grad produced by the dx kernel) into the same pass — saves a
+// separate torch reduce launch per call.
 template <typename T>
 __global__ void bdrl_bwd_dwdb_kernel(const T* __restrict__ dout,
                                      const T* __restrict__ xsum,
+                                     const T* __restrict__ dy,
                                      const float* __restrict__ mean,
                                      const float* __restrict__ rstd,
                                      float* __restrict__ dw32,
-                                     float* __restrict__ db32, long R, int H,
-                                     long rows_per_chunk) {
+                                     float* __restrict__ db32,
+                                     float* __restrict__ dbias32, long R,
+                                     int H, long rows_per_chunk) {
   const int col = blockIdx.x * blockDim.x + threadIdx.x;
   if (col >= H) return;
   const long r0 = blockIdx.y * rows_per_chunk;
   const long r1 = min(r0 + rows_per_chunk, R);
-  float dw = 0.f, db = 0.f;
+  float dw = 0.f, db = 0.f, dbias = 0.f;
   for (long r = r0; r < r1; ++r) {
     const float d = to_f32<T>(dout[r * H + col]);
     const float xh = (to_f32<T>(xsum[r * H + col]) - mean[r]) * rstd[r];
     dw += d * xh;
     db += d;
+    dbias += to_f32<T>(dy[r * H + col]);
   }
   atomicAdd(dw32 + col, dw);
   atomicAdd(db32 + col, db);
+  atomicAdd(dbias32 + col, dbias);
 }
 
 }  // namespace
@@ -183,10 +190,11 @@ std::vector<torch::Tensor> bias_dropout_residual_ln_bwd(
   auto dres = torch::empty_like(xsum);
   auto dw32 = torch::zeros({H}, xsum.options().dtype(torch::kFloat32));
   auto db32 = torch::zeros({H}, xsum.options().dtype(torch::kFloat32));
+  auto dbias32 = torch::zeros({H}, xsum.options().dtype(torch::kFloat32));
   auto stream = at::hip::getCurrentHIPStream();
   const bool drop = p > 0.0 && mask.numel() > 0;
-  const long chunks = std::max<long>(1, std::min<long>(32, R / 1024));
-  const long rows_per_chunk = (R + chunks - 1) / chunks;
+  const long rows_per_chunk = 16;
+  const long chunks = (R + rows_per_chunk - 1) / rows_per_chunk;
   DISPATCH_FLOAT_TYPES(xsum.scalar_type(), "bdrl_bwd", [&] {
     if (drop) {
       hipLaunchKernelGGL((bdrl_bwd_dx_kernel<scalar_t, true>), dim3(R),
@@ -213,12 +221,11 @@ std::vector<torch::Tensor> bias_dropout_residual_ln_bwd(
                        stream,
                        (const scalar_t*)dout.data_ptr(),
                        (const scalar_t*)xsum.data_ptr(),
+                       (const scalar_t*)dy.data_ptr(),
                        mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                       dw32.data_ptr<float>(), db32.data_ptr<float>(), R, H,
-                       rows_per_chunk);
+                       dw32.data_ptr<float>(), db32.data_ptr<float>(),
+                       dbias32.data_ptr<float>(), R, H, rows_per_chunk);
   });
   auto dt = lnw.scalar_type();
-  // dbias = column-sum of dy (post-dropout grad) — small reduce via torch
-  auto dbias = dy.view({R, (long)H}).sum(0);
-  return {dy, dbias, dres, dw32.to(dt), db32.to(dt)};
+  return {dy, dbias32.to(dt), dres, dw32.to(dt), db32.to(dt)};
 }

@@ -179,10 +179,9 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
   auto dw32 = torch::zeros({H}, x.options().dtype(torch::kFloat32));
   auto db32 = torch::zeros({H}, x.options().dtype(torch::kFloat32));
   auto stream = at::hip::getCurrentHIPStream();
-  const int chunks = std::min<long>(32, (R + 1023) / 1024) > 0
-                         ? std::min<long>(32, std::max<long>(1, R / 1024))
-                         : 1;
-  const int rows_per_chunk = (R + chunks - 1) / chunks;
+  // grid-fill: ~16 rows per chunk so (H/256)*chunks blocks cover 256 CUs
+  const int rows_per_chunk = 16;
+  const int chunks = (int)((R + rows_per_chunk - 1) / rows_per_chunk);
   DISPATCH_FLOAT_TYPES(x.scalar_type(), "layernorm_bwd", [&] {
     hipLaunchKernelGGL((ln_bwd_dx_kernel<scalar_t>), dim3(R), dim3(WAVE), 0,
                        stream,

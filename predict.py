@@ -64,7 +64,7 @@ def main():
         print(f"no checkpoints under {args.output_dir}; using random init")
         ckpts = [None]
     for c in ckpts:
-        model = build_model("bert-base", model_path=args.model_path)
+        model = build_model(args.model, model_path=args.model_path)
         if c is not None:
             load_checkpoint(model, c)
         model = model.to(device)

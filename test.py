@@ -26,7 +26,7 @@ from pdnlp_amd.utils import set_seed, load_checkpoint, strip_module_prefix
 
 def evaluate_checkpoint(path: str, args: Args):
     device = torch.device("cuda:0" if torch.cuda.is_available() else "cpu")
-    model = build_model("bert-base", model_path=args.model_path)
+    model = build_model(args.model, model_path=args.model_path)
     if os.path.isdir(path) and glob.glob(os.path.join(path, "zero_shard_r*.pt")):
         sd = consolidate_zero_checkpoint(path)
         model.load_state_dict({k: v for k, v in sd.items()}, strict=False)

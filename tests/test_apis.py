@@ -1,0 +1,131 @@
+"""CPU tests for the high-level APIs (Accelerator / HF-style Trainer /
+Fabric) and the CLI entrypoints."""
+
+import json
+import os
+import subprocess
+import sys
+
+import pytest
+import torch
+from torch.utils.data import DataLoader
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+pytestmark = pytest.mark.timeout(600)
+
+
+def _tiny_dataset(n=16, vocab=512):
+    from pdnlp_amd.data import SyntheticClsDataset
+    return SyntheticClsDataset(n, seq_len=16, vocab_size=vocab)
+
+
+def test_accelerator_single_process(tiny_cfg, tmp_path):
+    from pdnlp_amd.data import Collate
+    from pdnlp_amd.engine import Accelerator
+    from pdnlp_amd.models import BertForSequenceClassification
+    from pdnlp_amd.ops.adamw import build_optimizer
+
+    acc = Accelerator()
+    model = BertForSequenceClassification(tiny_cfg)
+    opt = build_optimizer(model, lr=1e-4)
+    loader = DataLoader(_tiny_dataset(), batch_size=4,
+                        collate_fn=Collate(None, 16))
+    model, opt, loader = acc.prepare(model, opt, loader)
+    for batch in loader:
+        out = model(batch["input_ids"], batch["attention_mask"],
+                    batch["token_type_ids"], batch["label"])
+        acc.backward(out.loss)
+        acc.step(opt, model)
+        opt.zero_grad(set_to_none=False)
+        break
+    assert acc.is_main_process
+    g = acc.gather(torch.ones(2, 3))
+    assert g.shape == (2, 3)
+
+
+def test_hf_style_trainer(tiny_cfg, tmp_path):
+    from pdnlp_amd.data import Collate
+    from pdnlp_amd.engine import HFStyleTrainer, TrainingArguments
+    from pdnlp_amd.models import BertForSequenceClassification
+
+    args = TrainingArguments(output_dir=str(tmp_path), eval_steps=2,
+                             per_device_train_batch_size=4,
+                             evaluation_strategy="steps", logging_steps=1)
+    trainer = HFStyleTrainer(
+        BertForSequenceClassification(tiny_cfg), args,
+        train_dataset=_tiny_dataset(16), eval_dataset=_tiny_dataset(8),
+        data_collator=Collate(None, 16, label_key="labels"))
+    res = trainer.train()
+    assert res["train_runtime_min"] > 0
+    metrics = trainer.evaluate()
+    assert "eval_accuracy" in metrics
+    preds = trainer.predict(_tiny_dataset(8))
+    assert preds.shape == (8, tiny_cfg.num_labels)
+    trainer.save_model(str(tmp_path / "m.pt"))
+    assert os.path.isfile(str(tmp_path / "m.pt"))
+
+
+def test_fabric_grad_accum(tiny_cfg):
+    from pdnlp_amd.data import Collate
+    from pdnlp_amd.engine import Fabric
+    from pdnlp_amd.models import BertForSequenceClassification
+    from pdnlp_amd.ops.adamw import build_optimizer
+
+    fabric = Fabric(devices=1, precision="32-true")
+    fabric.launch()
+    with fabric.init_module():
+        model = BertForSequenceClassification(tiny_cfg)
+    opt = build_optimizer(model, lr=1e-4)
+    model, opt = fabric.setup(model, opt)
+    loader = fabric.setup_dataloaders(
+        DataLoader(_tiny_dataset(), batch_size=4, collate_fn=Collate(None, 16)))
+    for i, batch in enumerate(loader):
+        out = model(batch["input_ids"], batch["attention_mask"],
+                    batch["token_type_ids"], batch["label"])
+        fabric.backward(out.loss / 2)
+        if (i + 1) % 2 == 0:
+            fabric.optimizer_step(opt, model)
+            opt.zero_grad(set_to_none=False)
+    assert out.loss.item() > 0
+
+
+def _run(script, *extra):
+    cmd = [sys.executable, os.path.join(REPO, script),
+           "--model", "tiny", "--data-limit", "48", "--max-seq-len", "16",
+           "--train-batch-size", "8", "--eval-step", "3",
+           "--num-workers", "0", *extra]
+    return subprocess.run(cmd, cwd=REPO, capture_output=True, text=True,
+                          timeout=420)
+
+
+@pytest.mark.parametrize("script", ["single-gpu-cls.py",
+                                    "multi-gpu-distributed-mp-cls.py"])
+def test_cli_entrypoints_cpu(script, tmp_path):
+    r = _run(script, "--ckpt-path", str(tmp_path / "m.pt"))
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert "【train】" in r.stdout
+    assert "耗时" in r.stdout
+    assert os.path.isfile(str(tmp_path / "m.pt"))
+
+
+def test_cli_predict_and_test(tmp_path):
+    # train a tiny ckpt, then audit it with test.py and predict.py
+    r = _run("single-gpu-cls.py", "--ckpt-path", str(tmp_path / "m.pt"))
+    assert r.returncode == 0, r.stderr[-2000:]
+    r2 = subprocess.run(
+        [sys.executable, os.path.join(REPO, "test.py"), "--ckpt",
+         str(tmp_path / "m.pt"), "--model", "tiny", "--data-limit", "48",
+         "--max-seq-len", "16", "--num-workers", "0"],
+        cwd=REPO, capture_output=True, text=True, timeout=420)
+    assert r2.returncode == 0, r2.stderr[-2000:]
+    assert "acc" in r2.stdout
+
+
+def test_metrics_jsonl_schema(tmp_path):
+    r = _run("single-gpu-cls.py", "--ckpt-path", str(tmp_path / "m.pt"),
+             "--metrics-jsonl", str(tmp_path / "met.jsonl"))
+    assert r.returncode == 0, r.stderr[-2000:]
+    lines = [json.loads(l) for l in open(tmp_path / "met.jsonl")]
+    train_lines = [l for l in lines if l.get("phase") == "train"]
+    assert train_lines and all("loss" in l and "step" in l for l in train_lines)

@@ -69,7 +69,8 @@ def test_bert_base_gpu_train_step_decreases_loss():
         opt.step()
         losses.append(out.loss.item())
     assert all(l == l for l in losses), f"NaN in {losses}"
-    assert losses[-1] < losses[0] * 0.7, losses
+    # fixed batch, lr 5e-5, dropout on: expect a clear downward trend
+    assert losses[-1] < losses[0] - 0.05, losses
 
 
 def test_bert_large_seq512_gpu_step():
