@@ -75,11 +75,46 @@ class BertEmbeddings(nn.Module):
 
 
 class BertSelfAttention(nn.Module):
+    """Q/K/V projections stored as ONE fused [3H, H] parameter (single MFMA
+    GEMM per layer, no per-step torch.cat, gradients land fused) while the
+    state dict still presents the HF keys ``query.weight``/``key.bias``/…
+    via save/load hooks."""
+
     def __init__(self, cfg: BertConfig):
         super().__init__()
-        self.query = nn.Linear(cfg.hidden_size, cfg.hidden_size)
-        self.key = nn.Linear(cfg.hidden_size, cfg.hidden_size)
-        self.value = nn.Linear(cfg.hidden_size, cfg.hidden_size)
+        H = cfg.hidden_size
+        self.hidden = H
+        self.qkv_weight = nn.Parameter(torch.empty(3 * H, H))
+        self.qkv_bias = nn.Parameter(torch.zeros(3 * H))
+        nn.init.normal_(self.qkv_weight, std=cfg.initializer_range)
+        self._register_state_dict_hook(_qkv_save_hook)
+        self._register_load_state_dict_pre_hook(self._qkv_load_hook,
+                                                with_module=False)
+
+    def _qkv_load_hook(self, state_dict, prefix, *args):
+        H = self.hidden
+        names = [("query", 0), ("key", 1), ("value", 2)]
+        if prefix + "qkv_weight" in state_dict:
+            return
+        ws, bs = [], []
+        for name, _ in names:
+            wk, bk = prefix + name + ".weight", prefix + name + ".bias"
+            if wk not in state_dict:
+                return  # let load_state_dict report what is missing
+            ws.append(state_dict.pop(wk))
+            bs.append(state_dict.pop(bk))
+        state_dict[prefix + "qkv_weight"] = torch.cat(ws, dim=0)
+        state_dict[prefix + "qkv_bias"] = torch.cat(bs, dim=0)
+
+
+def _qkv_save_hook(module, state_dict, prefix, local_metadata):
+    H = module.hidden
+    w = state_dict.pop(prefix + "qkv_weight")
+    b = state_dict.pop(prefix + "qkv_bias")
+    for i, name in enumerate(("query", "key", "value")):
+        state_dict[prefix + name + ".weight"] = w[i * H:(i + 1) * H]
+        state_dict[prefix + name + ".bias"] = b[i * H:(i + 1) * H]
+    return state_dict
 
 
 class BertSelfOutput(nn.Module):
@@ -124,9 +159,7 @@ class BertLayer(nn.Module):
         nh, hd = cfg.num_attention_heads, cfg.head_dim
         a = self.attention.self
         # fused QKV projection: one [H, 3H] GEMM (K2)
-        wqkv = torch.cat([a.query.weight, a.key.weight, a.value.weight], dim=0)
-        bqkv = torch.cat([a.query.bias, a.key.bias, a.value.bias], dim=0)
-        qkv = ops.linear(h, wqkv, bqkv)
+        qkv = ops.linear(h, a.qkv_weight, a.qkv_bias)
         q, k, v = qkv.split(H, dim=-1)
         q = q.view(B, S, nh, hd).transpose(1, 2)
         k = k.view(B, S, nh, hd).transpose(1, 2)
