@@ -47,6 +47,9 @@ def parse_args():
     p.add_argument("--bucket-cap-mb", type=float, default=50.0)
     p.add_argument("--zero", action="store_true", help="ZeRO mode instead of DDP")
     p.add_argument("--no-ddp-overlap", action="store_true")
+    p.add_argument("--graph", action="store_true",
+                   help="hipGraph-capture the fwd+bwd of the step "
+                        "(single-GPU; optimizer stays eager)")
     return p.parse_args()
 
 
@@ -107,13 +110,16 @@ def main():
     type_ids = torch.zeros_like(ids)
     labels = torch.randint(0, 6, (ns.batch_size,), generator=g).to(device)
 
-    def step():
+    def fwd_bwd():
         out = wrapped(input_ids=ids, attention_mask=mask,
                       token_type_ids=type_ids, labels=labels)
         loss = out.loss
         if scaler is not None:
             loss = scaler.scale(loss)
         loss.backward()
+        return out.loss
+
+    def opt_step():
         if isinstance(wrapped, DistributedDataParallel):
             wrapped.finalize_backward()
         if scaler is not None:
@@ -125,7 +131,33 @@ def main():
             wrapped.zero_grad_buffers()
         else:
             optimizer.zero_grad(set_to_none=False)
-        return out.loss
+
+    graph = None
+    if ns.graph and use_cuda and world == 1:
+        from pdnlp_amd.ops import reseed_dropout
+        # warm up allocator + autograd on a side stream, then capture
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(3):
+                fwd_bwd()
+                opt_step()
+        torch.cuda.current_stream().wait_stream(s)
+        graph = torch.cuda.CUDAGraph()
+        optimizer.zero_grad(set_to_none=False)
+        with torch.cuda.graph(graph):
+            static_loss = fwd_bwd()
+
+        def step():
+            reseed_dropout()
+            graph.replay()
+            opt_step()
+            return static_loss
+    else:
+        def step():
+            loss = fwd_bwd()
+            opt_step()
+            return loss
 
     for _ in range(ns.warmup):
         step()

@@ -81,5 +81,6 @@ from .functional import (  # noqa: F401,E402
     cross_entropy,
     bias_dropout_residual_layernorm,
     dropout,
+    reseed_dropout,
 )
 from .adamw import FusedAdamW, multi_tensor_adamw  # noqa: F401,E402

@@ -29,7 +29,8 @@ torch::Tensor gelu_bwd(torch::Tensor dy, torch::Tensor pre);
 torch::Tensor tanh_bwd(torch::Tensor dy, torch::Tensor pre);
 std::vector<torch::Tensor> bias_dropout_residual_ln_fwd(
     torch::Tensor y, torch::Tensor bias, torch::Tensor res, torch::Tensor lnw,
-    torch::Tensor lnb, double p, double eps, long seed);
+    torch::Tensor lnb, double p, double eps, torch::Tensor seed_buf,
+    long salt);
 std::vector<torch::Tensor> bias_dropout_residual_ln_bwd(
     torch::Tensor dout, torch::Tensor xsum, torch::Tensor mask,
     torch::Tensor lnw, torch::Tensor mean, torch::Tensor rstd, double p);

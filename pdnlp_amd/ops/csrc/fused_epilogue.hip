@@ -16,6 +16,9 @@
 
 namespace {
 
+// seed is read from DEVICE memory (seed_base) + a per-call-site salt so the
+// dropout mask changes across hipGraph replays (the host updates *seed_base
+// between replays; a kernel-arg seed would be frozen into the graph).
 template <typename T, bool DROP>
 __global__ void bdrl_fwd_kernel(const T* __restrict__ y,
                                 const T* __restrict__ bias,
@@ -26,7 +29,10 @@ __global__ void bdrl_fwd_kernel(const T* __restrict__ y,
                                 unsigned char* __restrict__ mask_out,
                                 float* __restrict__ mean_out,
                                 float* __restrict__ rstd_out, int H, float p,
-                                float eps, unsigned long long seed) {
+                                float eps,
+                                const unsigned long long* __restrict__ seed_base,
+                                unsigned long long salt) {
+  const unsigned long long seed = (DROP ? *seed_base : 0ull) + salt;
   const long row = blockIdx.x;
   const int lane = threadIdx.x & (WAVE - 1);
   const T* yr = y + row * H;
@@ -138,12 +144,17 @@ __global__ void bdrl_bwd_dwdb_kernel(const T* __restrict__ dout,
 
 std::vector<torch::Tensor> bias_dropout_residual_ln_fwd(
     torch::Tensor y, torch::Tensor bias, torch::Tensor res, torch::Tensor lnw,
-    torch::Tensor lnb, double p, double eps, long seed) {
+    torch::Tensor lnb, double p, double eps, torch::Tensor seed_buf,
+    long salt) {
   const int H = y.size(-1);
   const long R = y.numel() / H;
   auto out = torch::empty_like(y);
   auto xsum = torch::empty_like(y);
   const bool drop = p > 0.0;
+  TORCH_CHECK(!drop || (seed_buf.defined() && seed_buf.numel() >= 1 &&
+                        seed_buf.is_cuda() &&
+                        seed_buf.scalar_type() == torch::kLong),
+              "dropout needs a cuda int64 seed buffer");
   auto mask = drop
       ? torch::empty({R, (long)H}, y.options().dtype(torch::kUInt8))
       : torch::empty({0}, y.options().dtype(torch::kUInt8));
@@ -163,7 +174,9 @@ std::vector<torch::Tensor> bias_dropout_residual_ln_fwd(
                          (scalar_t*)xsum.data_ptr(),
                          mask.data_ptr<unsigned char>(),
                          mean.data_ptr<float>(), rstd.data_ptr<float>(), H,
-                         (float)p, (float)eps, (unsigned long long)seed);
+                         (float)p, (float)eps,
+                         (const unsigned long long*)seed_buf.data_ptr(),
+                         (unsigned long long)salt);
     } else {
       hipLaunchKernelGGL((bdrl_fwd_kernel<scalar_t, false>), dim3(R),
                          dim3(WAVE), 0, stream,
@@ -175,7 +188,7 @@ std::vector<torch::Tensor> bias_dropout_residual_ln_fwd(
                          (scalar_t*)out.data_ptr(),
                          (scalar_t*)xsum.data_ptr(), nullptr,
                          mean.data_ptr<float>(), rstd.data_ptr<float>(), H,
-                         (float)p, (float)eps, 0ull);
+                         (float)p, (float)eps, nullptr, 0ull);
     }
   });
   return {out, xsum, mask, mean, rstd};

@@ -235,14 +235,52 @@ class _FlashAttnFn(torch.autograd.Function):
 # Bias + dropout + residual + LayerNorm (K6/K8 epilogue)
 # --------------------------------------------------------------------------
 
+class _DropoutSeedState:
+    """Device-resident dropout seed (hipGraph-safe: the kernel reads the
+    seed from device memory, so ``reseed()`` between graph replays changes
+    the masks while the captured kernel args stay fixed). The per-call-site
+    salt keeps masks distinct within a step."""
+
+    def __init__(self):
+        self.buf = None
+        self.salt = 0
+
+    def get(self, device):
+        if self.buf is None or self.buf.device != device:
+            self.buf = torch.randint(0, 2**62, (1,), dtype=torch.int64,
+                                     device=device)
+        self.salt += 1
+        return self.buf, self.salt
+
+    def reseed(self, seed=None):
+        if self.buf is not None:
+            if seed is None:
+                seed = int(torch.randint(0, 2**62, (1,)).item())
+            self.buf.fill_(seed)
+
+
+_dropout_seed = _DropoutSeedState()
+
+
+def reseed_dropout(seed=None) -> None:
+    """Refresh the device dropout seed — call once per step when replaying
+    hipGraph-captured training steps."""
+    _dropout_seed.reseed(seed)
+
+
 class _BiasDropResLNFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, y, bias, residual, ln_w, ln_b, p, training, eps, seed):
+    def forward(ctx, y, bias, residual, ln_w, ln_b, p, training, eps):
+        p_eff = float(p if training else 0.0)
+        if p_eff > 0.0:
+            seed_buf, salt = _dropout_seed.get(y.device)
+        else:
+            seed_buf, salt = torch.Tensor(), 0
         out, xsum, mask, mean, rstd = ext().bias_dropout_residual_ln_fwd(
             y.contiguous(), bias, residual.contiguous(), ln_w, ln_b,
-            float(p if training else 0.0), eps, int(seed))
+            p_eff, eps, seed_buf, salt)
         ctx.save_for_backward(xsum, mask, ln_w, mean, rstd)
-        ctx.p = float(p if training else 0.0)
+        ctx.p = p_eff
         return out
 
     @staticmethod
@@ -260,9 +298,8 @@ def bias_dropout_residual_layernorm(
     """out = LN(dropout(y + bias) + residual) — the fused epilogue after the
     attention-output and FFN-down projections."""
     if hip_enabled(y):
-        seed = int(torch.randint(0, 2**31 - 1, (1,)).item()) if (p > 0 and training) else 0
         return _BiasDropResLNFn.apply(y, bias, residual, ln_w, ln_b, p,
-                                      training, eps, seed)
+                                      training, eps)
     h = y + bias
     if p > 0.0 and training:
         h = F.dropout(h, p=p, training=True)

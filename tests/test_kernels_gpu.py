@@ -147,7 +147,7 @@ def test_bdrl_no_dropout(dtype, rtol, atol):
     lnw = torch.randn(H, device=DEV, dtype=dtype)
     lnb = torch.randn(H, device=DEV, dtype=dtype)
     out, xsum, mask, mean, rstd = e.bias_dropout_residual_ln_fwd(
-        y, bias, res, lnw, lnb, 0.0, 1e-12, 0)
+        y, bias, res, lnw, lnb, 0.0, 1e-12, torch.Tensor(), 0)
     ref = F.layer_norm(y.float() + bias.float() + res.float(), (H,),
                        lnw.float(), lnb.float(), 1e-12)
     torch.testing.assert_close(out.float(), ref, rtol=rtol, atol=atol)
@@ -177,8 +177,9 @@ def test_bdrl_dropout_statistics():
     res = torch.zeros(R, H, device=DEV)
     lnw = torch.ones(H, device=DEV)
     lnb = torch.zeros(H, device=DEV)
+    seed = torch.tensor([1234], dtype=torch.int64, device=DEV)
     out, xsum, mask, mean, rstd = e.bias_dropout_residual_ln_fwd(
-        y, z, res, lnw, lnb, p, 1e-12, 1234)
+        y, z, res, lnw, lnb, p, 1e-12, seed, 7)
     keep_rate = mask.float().mean().item()
     assert abs(keep_rate - (1 - p)) < 5e-3, keep_rate
     # kept elements are scaled by 1/(1-p) before the residual add
@@ -186,10 +187,15 @@ def test_bdrl_dropout_statistics():
     torch.testing.assert_close(xsum[kept], y[kept] / (1 - p),
                                rtol=1e-5, atol=1e-5)
     assert (xsum[~kept] == 0).all()
-    # determinism in seed
+    # determinism in (device seed, salt)
     out2, xsum2, mask2, _, _ = e.bias_dropout_residual_ln_fwd(
-        y, z, res, lnw, lnb, p, 1e-12, 1234)
+        y, z, res, lnw, lnb, p, 1e-12, seed, 7)
     assert torch.equal(mask, mask2)
+    # updating the DEVICE seed changes the mask (hipGraph replay contract)
+    seed.fill_(99)
+    _, _, mask3, _, _ = e.bias_dropout_residual_ln_fwd(
+        y, z, res, lnw, lnb, p, 1e-12, seed, 7)
+    assert not torch.equal(mask, mask3)
 
 
 # ---------------------------------------------------------------------- gemm
