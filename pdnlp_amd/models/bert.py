@@ -158,15 +158,11 @@ class BertLayer(nn.Module):
         B, S, H = h.shape
         nh, hd = cfg.num_attention_heads, cfg.head_dim
         a = self.attention.self
-        # fused QKV projection: one [H, 3H] GEMM (K2)
+        # fused QKV projection: one [H, 3H] GEMM (K2), then fused flash
+        # attention straight on the packed projection (K3-K5 + K16)
         qkv = ops.linear(h, a.qkv_weight, a.qkv_bias)
-        q, k, v = qkv.split(H, dim=-1)
-        q = q.view(B, S, nh, hd).transpose(1, 2)
-        k = k.view(B, S, nh, hd).transpose(1, 2)
-        v = v.view(B, S, nh, hd).transpose(1, 2)
-        ctx = ops.attention(q, k, v, attn_mask,
-                            cfg.attention_probs_dropout_prob, training)
-        ctx = ctx.transpose(1, 2).reshape(B, S, H)
+        ctx = ops.attention_packed(qkv, attn_mask, nh,
+                                   cfg.attention_probs_dropout_prob, training)
         # attention output projection + fused bias/dropout/residual/LN (K6)
         ao = self.attention.output
         proj = ops.linear(ctx, ao.dense.weight, None)

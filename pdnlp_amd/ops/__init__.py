@@ -77,6 +77,7 @@ from .functional import (  # noqa: F401,E402
     linear,
     bias_gelu,
     attention,
+    attention_packed,
     masked_softmax,
     cross_entropy,
     bias_dropout_residual_layernorm,

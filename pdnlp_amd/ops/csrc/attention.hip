@@ -1,0 +1,669 @@
+// Fused flash-style attention for CDNA4/gfx950 — SURVEY.md K3+K4+K5 in one
+// kernel (reference: the cuBLAS batched QK^T / softmax / PV inside HF BERT,
+// multi-gpu-distributed-cls.py:132-136), plus its full backward.
+//
+// MI355X-first design decisions:
+//  - Operates DIRECTLY on the packed QKV projection output [B, S, 3H]
+//    (rows of 3H elements; head h's q/k/v at columns h*64 / H+h*64 / 2H+h*64)
+//    and writes a [B, S, H] context tensor — no split/transpose/contiguous
+//    copies on either side of the kernel, and backward emits dqkv in the
+//    same packed layout for the fused-QKV GEMM backward.
+//  - Online softmax (flash-style): S×S scores never hit HBM; handles
+//    S=128 (BERT-base) through S=512 (BERT-large) with the same code.
+//  - MFMA v_mfma_f32_16x16x32_bf16/f16 everywhere; fp32 accumulation.
+//    Block = 4 waves × 16 rows = 64-row tile; KV streamed in 64-key tiles,
+//    double-buffered through LDS via global_load_lds.
+//  - Attention-probability dropout is fused: mask is a counter-based hash of
+//    (device-resident seed + salt, element index), recomputed bit-exactly in
+//    backward — nothing saved but (o, lse). hipGraph-safe (seed read from
+//    device memory; host reseeds between replays).
+//  - Backward: two atomics-free passes. dQ pass owns 64-row blocks; dK/dV
+//    pass owns 64-key blocks and computes S^T = K@Q^T so every GEMM is the
+//    native MFMA A@B^T form. Transposed B-operands (V^T, K^T, Q^T, dO^T) are
+//    gathered from the row-major LDS image with per-lane b16 reads.
+//
+// LDS image convention (shared with gemm.hip): [64 rows][128 bytes], byte
+// offset XOR-swizzled by ((row & 7) << 4); staged by global_load_lds with the
+// swizzle pre-applied to the *source* address (writes are lane-linear).
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "common.h"
+
+namespace {
+
+typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
+typedef _Float16 f16x8 __attribute__((ext_vector_type(8)));
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+
+constexpr int NTHREADS = 256;  // 4 waves
+
+template <typename V8>
+__device__ __forceinline__ f32x4 mfma16(V8 a, V8 b, f32x4 c) {
+  if constexpr (std::is_same<V8, bf16x8>::value)
+    return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+  else
+    return __builtin_amdgcn_mfma_f32_16x16x32_f16(a, b, c, 0, 0, 0);
+}
+
+template <typename T>
+__device__ __forceinline__ unsigned short f32_bits16(float v) {
+  T t = from_f32<T>(v);
+  return *reinterpret_cast<unsigned short*>(&t);
+}
+
+// stage a 64-row x 64-elem (128 B) tile into the swizzled LDS image.
+// `ld` = row stride in elements; rows clamped to max_row (reads only).
+template <typename T>
+__device__ __forceinline__ void stage64(const T* __restrict__ src, long ld,
+                                        long row0, long max_row, char* lds) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x >> 6;
+  const int sub_row = lane >> 3;            // 0..7
+  const int piece = lane & 7;               // 16B piece of the 128B row
+  const int kbyte = (piece * 16) ^ (sub_row << 4);
+#pragma unroll
+  for (int c = 0; c < 2; ++c) {
+    const int r = (wid * 2 + c) * 8 + sub_row;
+    long gr = row0 + r;
+    gr = gr < max_row ? gr : max_row - 1;
+    const char* gp = (const char*)(src + gr * ld) + kbyte;
+    char* lp = lds + (long)(wid * 2 + c) * 8 * 128;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)gp,
+        (__attribute__((address_space(3))) unsigned int*)lp, 16, 0, 0);
+  }
+}
+
+// A/B-operand fragment: lane l holds image[row0 + (l&15)][(l>>4)*8 + i],
+// i=0..7 within the ks-th 32-element K-subtile — one ds_read_b128.
+template <typename V8>
+__device__ __forceinline__ V8 read_frag16(const char* lds, int row0, int ks) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int row = row0 + (lane & 15);
+  const int colbyte = (ks * 64 + (lane >> 4) * 16) ^ ((row & 7) << 4);
+  return *reinterpret_cast<const V8*>(lds + row * 128 + colbyte);
+}
+
+// Transposed fragment: B-operand rows are the image's COLUMNS. Lane l holds
+// image[key0 + (l>>4)*8 + i][d0 + (l&15)] — eight b16 gathers per fragment.
+template <typename V8>
+__device__ __forceinline__ V8 read_frag_t(const char* lds, int key0, int d0) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int dbyte = (d0 + (lane & 15)) * 2;
+  V8 out;
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    const int row = key0 + (lane >> 4) * 8 + i;
+    reinterpret_cast<unsigned short*>(&out)[i] =
+        *reinterpret_cast<const unsigned short*>(
+            lds + row * 128 + (dbyte ^ ((row & 7) << 4)));
+  }
+  return out;
+}
+
+// 16-lane (quarter-wave) butterfly reductions — score rows live across the
+// 16 lanes that share lane>>4.
+__device__ __forceinline__ float qmax(float x) {
+#pragma unroll
+  for (int off = 1; off < 16; off <<= 1) x = fmaxf(x, __shfl_xor(x, off, WAVE));
+  return x;
+}
+__device__ __forceinline__ float qsum(float x) {
+#pragma unroll
+  for (int off = 1; off < 16; off <<= 1) x += __shfl_xor(x, off, WAVE);
+  return x;
+}
+
+// ---------------------------------------------------------------------------
+// forward
+// ---------------------------------------------------------------------------
+
+template <typename T, typename V8, bool HAS_MASK, bool DROP>
+__global__ __launch_bounds__(NTHREADS)
+void fa_fwd_kernel(const T* __restrict__ qkv, const T* __restrict__ mask,
+                   T* __restrict__ o, float* __restrict__ lse,
+                   const unsigned long long* __restrict__ seed_base,
+                   unsigned long long salt, float scale, float p,
+                   int nh, int S) {
+  const int bh = blockIdx.y;
+  const long b = bh / nh, h = bh % nh;
+  const long rb = blockIdx.x;
+  const long H = (long)nh * 64, ld = 3 * H;
+  const T* qbase = qkv + b * S * ld + h * 64;
+  const T* kbase = qbase + H;
+  const T* vbase = qbase + 2 * H;
+  T* obase = o + b * S * H + h * 64;
+
+  __shared__ __attribute__((aligned(16))) char q_lds[64 * 128];
+  __shared__ __attribute__((aligned(16))) char k_lds[2][64 * 128];
+  __shared__ __attribute__((aligned(16))) char v_lds[2][64 * 128];
+  __shared__ __attribute__((aligned(16))) char p_lds[64 * 128];
+
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x >> 6;
+  const int wr = wid * 16;
+  const unsigned long long seed = DROP ? (*seed_base + salt) : 0ull;
+  const float inv_keep = DROP ? 1.f / (1.f - p) : 1.f;
+
+  stage64<T>(qbase, ld, rb * 64, S, q_lds);
+  stage64<T>(kbase, ld, 0, S, k_lds[0]);
+  stage64<T>(vbase, ld, 0, S, v_lds[0]);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+
+  float m[4], l[4];
+  f32x4 acc_o[4] = {};
+#pragma unroll
+  for (int r = 0; r < 4; ++r) { m[r] = -INFINITY; l[r] = 0.f; }
+
+  const int nt = S / 64;
+  int cur = 0;
+  for (int t = 0; t < nt; ++t) {
+    if (t + 1 < nt) {
+      stage64<T>(kbase, ld, (long)(t + 1) * 64, S, k_lds[cur ^ 1]);
+      stage64<T>(vbase, ld, (long)(t + 1) * 64, S, v_lds[cur ^ 1]);
+    }
+    f32x4 acc_s[4] = {};
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      V8 a = read_frag16<V8>(q_lds, wr, ks);
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        acc_s[j] =
+            mfma16<V8>(a, read_frag16<V8>(k_lds[cur], j * 16, ks), acc_s[j]);
+    }
+    float mv[4];
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+      mv[j] = HAS_MASK
+                  ? to_f32<T>(mask[b * S + t * 64 + j * 16 + (lane & 15)])
+                  : 0.f;
+    float s[4][4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      float x = -INFINITY;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        s[j][r] = acc_s[j][r] * scale + mv[j];
+        x = fmaxf(x, s[j][r]);
+      }
+      const float mn = fmaxf(m[r], qmax(x));
+      const float alpha = expf(m[r] - mn);   // first tile: exp(-inf) = 0
+      m[r] = mn;
+      float sum = 0.f;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        s[j][r] = expf(s[j][r] - mn);
+        sum += s[j][r];
+      }
+      l[r] = l[r] * alpha + qsum(sum);
+#pragma unroll
+      for (int jd = 0; jd < 4; ++jd) acc_o[jd][r] *= alpha;
+    }
+    // dropout + P tile into LDS (own wave's 16 rows only)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int prow = wr + (lane >> 4) * 4 + r;
+      const unsigned long long grow = (unsigned long long)bh * S + rb * 64 + prow;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const int kcol = j * 16 + (lane & 15);
+        float pv = s[j][r];
+        if (DROP) {
+          const unsigned int rr = hash_rng(seed, grow * S + t * 64 + kcol);
+          pv = (rr * 2.3283064365386963e-10f) >= p ? pv * inv_keep : 0.f;
+        }
+        *reinterpret_cast<unsigned short*>(
+            p_lds + prow * 128 + ((kcol * 2) ^ ((prow & 7) << 4))) =
+            f32_bits16<T>(pv);
+      }
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      V8 a = read_frag16<V8>(p_lds, wr, ks);
+#pragma unroll
+      for (int jd = 0; jd < 4; ++jd)
+        acc_o[jd] = mfma16<V8>(a, read_frag_t<V8>(v_lds[cur], ks * 32, jd * 16),
+                               acc_o[jd]);
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  const int ccol = lane & 15;
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int prow = wr + (lane >> 4) * 4 + r;
+    const long grow = rb * 64 + prow;
+    const float invl = l[r] > 0.f ? 1.f / l[r] : 0.f;
+#pragma unroll
+    for (int jd = 0; jd < 4; ++jd)
+      obase[grow * H + jd * 16 + ccol] = from_f32<T>(acc_o[jd][r] * invl);
+    if (ccol == 0)
+      lse[(long)bh * S + grow] = m[r] + logf(fmaxf(l[r], 1e-30f));
+  }
+}
+
+// ---------------------------------------------------------------------------
+// backward: Dvec = rowsum(dO * O) — one wave per (b, s, h) row, D=64 lanes
+// ---------------------------------------------------------------------------
+
+template <typename T>
+__global__ __launch_bounds__(NTHREADS)
+void fa_bwd_pre_kernel(const T* __restrict__ dout, const T* __restrict__ o,
+                       float* __restrict__ dvec, int nh, int S, long total) {
+  const long idx = (long)blockIdx.x * 4 + (threadIdx.x >> 6);
+  if (idx >= total) return;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const long H = (long)nh * 64;
+  const long bh = idx / S, srow = idx % S;
+  const long b = bh / nh, h = bh % nh;
+  const long off = (b * S + srow) * H + h * 64 + lane;
+  const float v = to_f32<T>(dout[off]) * to_f32<T>(o[off]);
+  const float sum = wave_sum(v);
+  if (lane == 0) dvec[idx] = sum;
+}
+
+// ---------------------------------------------------------------------------
+// backward dQ: grid over 64-row blocks; recompute P from lse, stream K/V
+// ---------------------------------------------------------------------------
+
+template <typename T, typename V8, bool HAS_MASK, bool DROP>
+__global__ __launch_bounds__(NTHREADS)
+void fa_bwd_dq_kernel(const T* __restrict__ dout, const T* __restrict__ qkv,
+                      const float* __restrict__ lse,
+                      const float* __restrict__ dvec,
+                      const T* __restrict__ mask, T* __restrict__ dqkv,
+                      const unsigned long long* __restrict__ seed_base,
+                      unsigned long long salt, float scale, float p,
+                      int nh, int S) {
+  const int bh = blockIdx.y;
+  const long b = bh / nh, h = bh % nh;
+  const long rb = blockIdx.x;
+  const long H = (long)nh * 64, ld = 3 * H;
+  const T* qbase = qkv + b * S * ld + h * 64;
+  const T* kbase = qbase + H;
+  const T* vbase = qbase + 2 * H;
+  const T* dobase = dout + b * S * H + h * 64;
+  T* dqbase = dqkv + b * S * ld + h * 64;
+
+  __shared__ __attribute__((aligned(16))) char q_lds[64 * 128];
+  __shared__ __attribute__((aligned(16))) char do_lds[64 * 128];
+  __shared__ __attribute__((aligned(16))) char k_lds[2][64 * 128];
+  __shared__ __attribute__((aligned(16))) char v_lds[2][64 * 128];
+  __shared__ __attribute__((aligned(16))) char ds_lds[64 * 128];
+
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x >> 6;
+  const int wr = wid * 16;
+  const unsigned long long seed = DROP ? (*seed_base + salt) : 0ull;
+  const float inv_keep = DROP ? 1.f / (1.f - p) : 1.f;
+
+  stage64<T>(qbase, ld, rb * 64, S, q_lds);
+  stage64<T>(dobase, H, rb * 64, S, do_lds);
+  stage64<T>(kbase, ld, 0, S, k_lds[0]);
+  stage64<T>(vbase, ld, 0, S, v_lds[0]);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+
+  float lse_r[4], dvec_r[4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const long grow = rb * 64 + wr + (lane >> 4) * 4 + r;
+    lse_r[r] = lse[(long)bh * S + grow];
+    dvec_r[r] = dvec[(long)bh * S + grow];
+  }
+
+  f32x4 acc_dq[4] = {};
+  const int nt = S / 64;
+  int cur = 0;
+  for (int t = 0; t < nt; ++t) {
+    if (t + 1 < nt) {
+      stage64<T>(kbase, ld, (long)(t + 1) * 64, S, k_lds[cur ^ 1]);
+      stage64<T>(vbase, ld, (long)(t + 1) * 64, S, v_lds[cur ^ 1]);
+    }
+    f32x4 acc_s[4] = {}, acc_dp[4] = {};
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      V8 aq = read_frag16<V8>(q_lds, wr, ks);
+      V8 ad = read_frag16<V8>(do_lds, wr, ks);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        acc_s[j] =
+            mfma16<V8>(aq, read_frag16<V8>(k_lds[cur], j * 16, ks), acc_s[j]);
+        acc_dp[j] =
+            mfma16<V8>(ad, read_frag16<V8>(v_lds[cur], j * 16, ks), acc_dp[j]);
+      }
+    }
+    float mv[4];
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+      mv[j] = HAS_MASK
+                  ? to_f32<T>(mask[b * S + t * 64 + j * 16 + (lane & 15)])
+                  : 0.f;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int prow = wr + (lane >> 4) * 4 + r;
+      const unsigned long long grow = (unsigned long long)bh * S + rb * 64 + prow;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const int kcol = j * 16 + (lane & 15);
+        const float pv = expf(acc_s[j][r] * scale + mv[j] - lse_r[r]);
+        float dp = acc_dp[j][r];
+        if (DROP) {
+          const unsigned int rr = hash_rng(seed, grow * S + t * 64 + kcol);
+          dp = (rr * 2.3283064365386963e-10f) >= p ? dp * inv_keep : 0.f;
+        }
+        const float dsv = pv * (dp - dvec_r[r]) * scale;
+        *reinterpret_cast<unsigned short*>(
+            ds_lds + prow * 128 + ((kcol * 2) ^ ((prow & 7) << 4))) =
+            f32_bits16<T>(dsv);
+      }
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      V8 a = read_frag16<V8>(ds_lds, wr, ks);
+#pragma unroll
+      for (int jd = 0; jd < 4; ++jd)
+        acc_dq[jd] = mfma16<V8>(
+            a, read_frag_t<V8>(k_lds[cur], ks * 32, jd * 16), acc_dq[jd]);
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  const int ccol = lane & 15;
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const long grow = rb * 64 + wr + (lane >> 4) * 4 + r;
+#pragma unroll
+    for (int jd = 0; jd < 4; ++jd)
+      dqbase[grow * ld + jd * 16 + ccol] = from_f32<T>(acc_dq[jd][r]);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// backward dK/dV: grid over 64-key blocks; S^T = K@Q^T so all GEMMs are A@B^T
+// ---------------------------------------------------------------------------
+
+template <typename T, typename V8, bool HAS_MASK, bool DROP>
+__global__ __launch_bounds__(NTHREADS)
+void fa_bwd_dkv_kernel(const T* __restrict__ dout, const T* __restrict__ qkv,
+                       const float* __restrict__ lse,
+                       const float* __restrict__ dvec,
+                       const T* __restrict__ mask, T* __restrict__ dqkv,
+                       const unsigned long long* __restrict__ seed_base,
+                       unsigned long long salt, float scale, float p,
+                       int nh, int S) {
+  const int bh = blockIdx.y;
+  const long b = bh / nh, h = bh % nh;
+  const long kb = blockIdx.x;
+  const long H = (long)nh * 64, ld = 3 * H;
+  const T* qbase = qkv + b * S * ld + h * 64;
+  const T* kbase = qbase + H;
+  const T* vbase = qbase + 2 * H;
+  const T* dobase = dout + b * S * H + h * 64;
+  T* dkbase = dqkv + b * S * ld + H + h * 64;
+  T* dvbase = dqkv + b * S * ld + 2 * H + h * 64;
+
+  __shared__ __attribute__((aligned(16))) char k_lds[64 * 128];
+  __shared__ __attribute__((aligned(16))) char v_lds[64 * 128];
+  __shared__ __attribute__((aligned(16))) char q_lds[2][64 * 128];
+  __shared__ __attribute__((aligned(16))) char do_lds[2][64 * 128];
+  __shared__ __attribute__((aligned(16))) char pds_lds[64 * 128];
+
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x >> 6;
+  const int kr = wid * 16;
+  const unsigned long long seed = DROP ? (*seed_base + salt) : 0ull;
+  const float inv_keep = DROP ? 1.f / (1.f - p) : 1.f;
+
+  stage64<T>(kbase, ld, kb * 64, S, k_lds);
+  stage64<T>(vbase, ld, kb * 64, S, v_lds);
+  stage64<T>(qbase, ld, 0, S, q_lds[0]);
+  stage64<T>(dobase, H, 0, S, do_lds[0]);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+
+  float mv[4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const long gkey = kb * 64 + kr + (lane >> 4) * 4 + r;
+    mv[r] = HAS_MASK ? to_f32<T>(mask[b * S + gkey]) : 0.f;
+  }
+
+  f32x4 acc_dk[4] = {}, acc_dv[4] = {};
+  const int nt = S / 64;
+  int cur = 0;
+  for (int t = 0; t < nt; ++t) {
+    if (t + 1 < nt) {
+      stage64<T>(qbase, ld, (long)(t + 1) * 64, S, q_lds[cur ^ 1]);
+      stage64<T>(dobase, H, (long)(t + 1) * 64, S, do_lds[cur ^ 1]);
+    }
+    // S^T = K @ Q^T and dP^T = V @ dO^T on this 64-query tile
+    f32x4 acc_st[4] = {}, acc_dpt[4] = {};
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      V8 ak = read_frag16<V8>(k_lds, kr, ks);
+      V8 av = read_frag16<V8>(v_lds, kr, ks);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        acc_st[j] = mfma16<V8>(
+            ak, read_frag16<V8>(q_lds[cur], j * 16, ks), acc_st[j]);
+        acc_dpt[j] = mfma16<V8>(
+            av, read_frag16<V8>(do_lds[cur], j * 16, ks), acc_dpt[j]);
+      }
+    }
+    float pt[4][4];
+    bool live[4][4];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const long qrow = (long)t * 64 + j * 16 + (lane & 15);
+      const float lse_j = lse[(long)bh * S + qrow];
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const long gkey = kb * 64 + kr + (lane >> 4) * 4 + r;
+        pt[j][r] = expf(acc_st[j][r] * scale + mv[r] - lse_j);
+        live[j][r] = true;
+        if (DROP) {
+          const unsigned int rr = hash_rng(
+              seed, ((unsigned long long)bh * S + qrow) * S + gkey);
+          live[j][r] = (rr * 2.3283064365386963e-10f) >= p;
+        }
+      }
+    }
+    // Pd^T tile -> LDS; dV += Pd^T @ dO
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int prow = kr + (lane >> 4) * 4 + r;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const int qcol = j * 16 + (lane & 15);
+        const float pv = live[j][r] ? pt[j][r] * inv_keep : 0.f;
+        *reinterpret_cast<unsigned short*>(
+            pds_lds + prow * 128 + ((qcol * 2) ^ ((prow & 7) << 4))) =
+            f32_bits16<T>(pv);
+      }
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      V8 a = read_frag16<V8>(pds_lds, kr, ks);
+#pragma unroll
+      for (int jd = 0; jd < 4; ++jd)
+        acc_dv[jd] = mfma16<V8>(
+            a, read_frag_t<V8>(do_lds[cur], ks * 32, jd * 16), acc_dv[jd]);
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    // dS^T tile -> LDS (same buffer); dK += dS^T @ Q
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int prow = kr + (lane >> 4) * 4 + r;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const long qrow = (long)t * 64 + j * 16 + (lane & 15);
+        const float dvec_j = dvec[(long)bh * S + qrow];
+        const float dpt = live[j][r] ? acc_dpt[j][r] * inv_keep : 0.f;
+        const float dsv = pt[j][r] * (dpt - dvec_j) * scale;
+        const int qcol = j * 16 + (lane & 15);
+        *reinterpret_cast<unsigned short*>(
+            pds_lds + prow * 128 + ((qcol * 2) ^ ((prow & 7) << 4))) =
+            f32_bits16<T>(dsv);
+      }
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      V8 a = read_frag16<V8>(pds_lds, kr, ks);
+#pragma unroll
+      for (int jd = 0; jd < 4; ++jd)
+        acc_dk[jd] = mfma16<V8>(
+            a, read_frag_t<V8>(q_lds[cur], ks * 32, jd * 16), acc_dk[jd]);
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  const int ccol = lane & 15;
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const long gkey = kb * 64 + kr + (lane >> 4) * 4 + r;
+#pragma unroll
+    for (int jd = 0; jd < 4; ++jd) {
+      dkbase[gkey * ld + jd * 16 + ccol] = from_f32<T>(acc_dk[jd][r]);
+      dvbase[gkey * ld + jd * 16 + ccol] = from_f32<T>(acc_dv[jd][r]);
+    }
+  }
+}
+
+}  // namespace
+
+// ---------------------------------------------------------------------------
+// host wrappers
+// ---------------------------------------------------------------------------
+
+static void check_fa_args(const torch::Tensor& qkv, const torch::Tensor& mask,
+                          long nh) {
+  TORCH_CHECK(qkv.is_cuda() && qkv.is_contiguous() && qkv.dim() == 3,
+              "flash_attn: qkv must be contiguous [B, S, 3H]");
+  TORCH_CHECK(qkv.scalar_type() == torch::kBFloat16 ||
+                  qkv.scalar_type() == torch::kHalf,
+              "flash_attn: bf16/fp16 only");
+  const long S = qkv.size(1);
+  TORCH_CHECK(qkv.size(2) == 3 * nh * 64,
+              "flash_attn: head_dim must be 64 and qkv last dim 3*nh*64");
+  TORCH_CHECK(S % 64 == 0, "flash_attn: S must be a multiple of 64");
+  if (mask.defined() && mask.numel() > 0) {
+    TORCH_CHECK(mask.is_contiguous() && mask.numel() == qkv.size(0) * S &&
+                    mask.scalar_type() == qkv.scalar_type(),
+                "flash_attn: mask must be contiguous [B,1,1,S] of qkv dtype");
+  }
+}
+
+// launch macros live at file scope (a #define cannot appear inside a macro
+// argument — DISPATCH_FLOAT_TYPES takes the body as one)
+#define FA_FWD(HM, DR)                                                         \
+  hipLaunchKernelGGL((fa_fwd_kernel<scalar_t, V8, HM, DR>), grid,              \
+                     dim3(NTHREADS), 0, stream,                                \
+                     (const scalar_t*)qkv.data_ptr(),                          \
+                     has_mask ? (const scalar_t*)mask.data_ptr() : nullptr,    \
+                     (scalar_t*)o.data_ptr(), (float*)lse.data_ptr(),          \
+                     seed_ptr, (unsigned long long)salt, (float)scale,         \
+                     (float)p, (int)nh, (int)S)
+
+#define FA_BWD(KERN, HM, DR)                                                   \
+  hipLaunchKernelGGL((KERN<scalar_t, V8, HM, DR>), grid, dim3(NTHREADS), 0,    \
+                     stream, (const scalar_t*)dout.data_ptr(),                 \
+                     (const scalar_t*)qkv.data_ptr(),                          \
+                     (const float*)lse.data_ptr(),                             \
+                     (const float*)dvec.data_ptr(),                            \
+                     has_mask ? (const scalar_t*)mask.data_ptr() : nullptr,    \
+                     (scalar_t*)dqkv.data_ptr(), seed_ptr,                     \
+                     (unsigned long long)salt, (float)scale, (float)p,         \
+                     (int)nh, (int)S)
+
+std::vector<torch::Tensor> flash_attn_qkv_fwd(torch::Tensor qkv,
+                                              torch::Tensor mask, long nh,
+                                              double scale, double p,
+                                              torch::Tensor seed_buf,
+                                              long salt) {
+  check_fa_args(qkv, mask, nh);
+  const long B = qkv.size(0), S = qkv.size(1), H = (long)nh * 64;
+  const bool has_mask = mask.defined() && mask.numel() > 0;
+  const bool drop = p > 0.0;
+  TORCH_CHECK(!drop || (seed_buf.defined() && seed_buf.is_cuda() &&
+                        seed_buf.scalar_type() == torch::kLong),
+              "flash_attn: dropout needs a cuda int64 seed buffer");
+  auto o = torch::empty({B, S, H}, qkv.options());
+  auto lse = torch::empty({B * nh, S}, qkv.options().dtype(torch::kFloat));
+  dim3 grid(S / 64, B * nh);
+  auto stream = at::hip::getCurrentHIPStream();
+  const auto* seed_ptr =
+      drop ? (const unsigned long long*)seed_buf.data_ptr() : nullptr;
+
+  DISPATCH_FLOAT_TYPES(qkv.scalar_type(), "flash_attn_qkv_fwd", [&] {
+    if constexpr (!std::is_same<scalar_t, float>::value) {
+      using V8 = std::conditional_t<
+          std::is_same<scalar_t, __hip_bfloat16>::value, bf16x8, f16x8>;
+      if (has_mask && drop) FA_FWD(true, true);
+      else if (has_mask) FA_FWD(true, false);
+      else if (drop) FA_FWD(false, true);
+      else FA_FWD(false, false);
+    } else {
+      TORCH_CHECK(false, "flash_attn: fp32 not supported");
+    }
+  });
+  return {o, lse};
+}
+
+torch::Tensor flash_attn_qkv_bwd(torch::Tensor dout, torch::Tensor qkv,
+                                 torch::Tensor o, torch::Tensor lse,
+                                 torch::Tensor mask, long nh, double scale,
+                                 double p, torch::Tensor seed_buf, long salt) {
+  check_fa_args(qkv, mask, nh);
+  const long B = qkv.size(0), S = qkv.size(1);
+  const bool has_mask = mask.defined() && mask.numel() > 0;
+  const bool drop = p > 0.0;
+  auto dqkv = torch::empty_like(qkv);
+  auto dvec = torch::empty({B * nh, S}, qkv.options().dtype(torch::kFloat));
+  auto stream = at::hip::getCurrentHIPStream();
+  const auto* seed_ptr =
+      drop ? (const unsigned long long*)seed_buf.data_ptr() : nullptr;
+  const long total = B * nh * S;
+
+  DISPATCH_FLOAT_TYPES(qkv.scalar_type(), "flash_attn_qkv_bwd", [&] {
+    if constexpr (!std::is_same<scalar_t, float>::value) {
+      using V8 = std::conditional_t<
+          std::is_same<scalar_t, __hip_bfloat16>::value, bf16x8, f16x8>;
+      hipLaunchKernelGGL((fa_bwd_pre_kernel<scalar_t>),
+                         dim3((total + 3) / 4), dim3(NTHREADS), 0, stream,
+                         (const scalar_t*)dout.data_ptr(),
+                         (const scalar_t*)o.data_ptr(),
+                         (float*)dvec.data_ptr(), (int)nh, (int)S, total);
+      dim3 grid(S / 64, B * nh);
+      if (has_mask && drop) {
+        FA_BWD(fa_bwd_dq_kernel, true, true);
+        FA_BWD(fa_bwd_dkv_kernel, true, true);
+      } else if (has_mask) {
+        FA_BWD(fa_bwd_dq_kernel, true, false);
+        FA_BWD(fa_bwd_dkv_kernel, true, false);
+      } else if (drop) {
+        FA_BWD(fa_bwd_dq_kernel, false, true);
+        FA_BWD(fa_bwd_dkv_kernel, false, true);
+      } else {
+        FA_BWD(fa_bwd_dq_kernel, false, false);
+        FA_BWD(fa_bwd_dkv_kernel, false, false);
+      }
+    } else {
+      TORCH_CHECK(false, "flash_attn: fp32 not supported");
+    }
+  });
+  return dqkv;
+}

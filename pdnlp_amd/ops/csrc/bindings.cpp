@@ -50,6 +50,15 @@ void multi_tensor_unscale(std::vector<torch::Tensor> grads,
                           torch::Tensor found_inf, double inv_scale);
 std::vector<torch::Tensor> gemm_nt_fwd(torch::Tensor A, torch::Tensor W,
                                        torch::Tensor bias, std::string act);
+std::vector<torch::Tensor> flash_attn_qkv_fwd(torch::Tensor qkv,
+                                              torch::Tensor mask, long nh,
+                                              double scale, double p,
+                                              torch::Tensor seed_buf,
+                                              long salt);
+torch::Tensor flash_attn_qkv_bwd(torch::Tensor dout, torch::Tensor qkv,
+                                 torch::Tensor o, torch::Tensor lse,
+                                 torch::Tensor mask, long nh, double scale,
+                                 double p, torch::Tensor seed_buf, long salt);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layernorm_fwd", &layernorm_fwd);
@@ -69,4 +78,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("multi_tensor_adamw", &multi_tensor_adamw);
   m.def("multi_tensor_unscale", &multi_tensor_unscale);
   m.def("gemm_nt_fwd", &gemm_nt_fwd);
+  m.def("flash_attn_qkv_fwd", &flash_attn_qkv_fwd);
+  m.def("flash_attn_qkv_bwd", &flash_attn_qkv_bwd);
 }

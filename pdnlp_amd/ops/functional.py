@@ -198,13 +198,12 @@ def masked_softmax(scores: torch.Tensor, mask: Optional[torch.Tensor],
 def attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
               mask: Optional[torch.Tensor], dropout_p: float = 0.0,
               training: bool = False) -> torch.Tensor:
-    """q,k,v: [B, H, S, D]; mask: additive [B, 1, 1, S]. Returns [B, H, S, D]."""
+    """q,k,v: [B, H, S, D]; mask: additive [B, 1, 1, S]. Returns [B, H, S, D].
+
+    Unfused composition (batched GEMMs + fused masked softmax) — the fallback
+    behind :func:`attention_packed` and the CPU numerics reference.
+    """
     scale = 1.0 / math.sqrt(q.shape[-1])
-    if hip_enabled(q) and getattr(ext(), "flash_attn_fwd", None) is not None \
-            and (dropout_p == 0.0 or not training) and q.shape[-1] in (64,) \
-            and q.dtype == torch.bfloat16:
-        return _FlashAttnFn.apply(q.contiguous(), k.contiguous(), v.contiguous(),
-                                  mask, scale)
     scores = torch.matmul(q, k.transpose(-1, -2))
     probs = masked_softmax(scores, mask, scale)
     if dropout_p > 0.0 and training:
@@ -212,23 +211,66 @@ def attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     return torch.matmul(probs, v)
 
 
-class _FlashAttnFn(torch.autograd.Function):
+class _FlashAttnQKVFn(torch.autograd.Function):
+    """Fused flash attention over the packed QKV projection [B, S, 3H] —
+    K3+K4+K5 (+ attention dropout K16) in one MFMA kernel; returns the
+    [B, S, H] context ready for the output projection. Dropout masks are
+    recomputed in backward from the device seed + salt (valid as long as
+    ``reseed_dropout`` is only called between steps, which is the step
+    contract everywhere in this framework)."""
+
     @staticmethod
-    def forward(ctx, q, k, v, mask, scale):
-        o, lse = ext().flash_attn_fwd(q, k, v,
-                                      mask.contiguous() if mask is not None
-                                      else torch.Tensor(), scale)
-        ctx.save_for_backward(q, k, v, o, lse,
-                              mask if mask is not None else torch.Tensor())
-        ctx.scale = scale
+    def forward(ctx, qkv, mask, nh, scale, p_drop):
+        if p_drop > 0.0:
+            seed_buf, salt = _dropout_seed.get(qkv.device)
+        else:
+            seed_buf, salt = torch.Tensor(), 0
+        mask_t = mask if mask is not None else torch.Tensor()
+        o, lse = ext().flash_attn_qkv_fwd(qkv, mask_t, nh, scale, p_drop,
+                                          seed_buf, salt)
+        ctx.save_for_backward(qkv, o, lse,
+                              mask if mask is not None else torch.Tensor(),
+                              seed_buf if p_drop > 0.0 else torch.Tensor())
+        ctx.nh, ctx.scale, ctx.p, ctx.salt = nh, scale, p_drop, salt
         return o
 
     @staticmethod
     def backward(ctx, do):
-        q, k, v, o, lse, mask = ctx.saved_tensors
-        dq, dk, dv = ext().flash_attn_bwd(do.contiguous(), q, k, v, o, lse,
-                                          mask, ctx.scale)
-        return dq, dk, dv, None, None
+        qkv, o, lse, mask, seed_buf = ctx.saved_tensors
+        dqkv = ext().flash_attn_qkv_bwd(do.contiguous(), qkv, o, lse, mask,
+                                        ctx.nh, ctx.scale, ctx.p, seed_buf,
+                                        ctx.salt)
+        return dqkv, None, None, None, None
+
+
+def attention_packed(qkv: torch.Tensor, mask: Optional[torch.Tensor],
+                     num_heads: int, dropout_p: float = 0.0,
+                     training: bool = False) -> torch.Tensor:
+    """Attention on the packed QKV projection output.
+
+    qkv: [B, S, 3H] (q | k | v along the last dim, H = num_heads*64);
+    mask: additive [B, 1, 1, S] in qkv's dtype. Returns context [B, S, H].
+    On a ROCm GPU with head_dim 64, S%64==0 and bf16/fp16 this runs the
+    fused MFMA flash kernel; otherwise it falls back to the split
+    batched-GEMM composition."""
+    B, S, H3 = qkv.shape
+    H = H3 // 3
+    hd = H // num_heads
+    p_eff = float(dropout_p if training else 0.0)
+    if hip_enabled(qkv) and getattr(ext(), "flash_attn_qkv_fwd", None) is not None \
+            and hd == 64 and S % 64 == 0 \
+            and qkv.dtype in (torch.bfloat16, torch.float16):
+        scale = 1.0 / math.sqrt(hd)
+        return _FlashAttnQKVFn.apply(
+            qkv.contiguous(),
+            mask.contiguous() if mask is not None else None,
+            num_heads, scale, p_eff)
+    q, k, v = qkv.split(H, dim=-1)
+    q = q.view(B, S, num_heads, hd).transpose(1, 2)
+    k = k.view(B, S, num_heads, hd).transpose(1, 2)
+    v = v.view(B, S, num_heads, hd).transpose(1, 2)
+    ctx = attention(q, k, v, mask, p_eff, training)
+    return ctx.transpose(1, 2).reshape(B, S, H)
 
 
 # --------------------------------------------------------------------------
