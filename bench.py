@@ -119,7 +119,7 @@ def main():
         loss.backward()
         return out.loss
 
-    def opt_step():
+    def opt_step(zero: bool = True):
         if isinstance(wrapped, DistributedDataParallel):
             wrapped.finalize_backward()
         if scaler is not None:
@@ -129,8 +129,10 @@ def main():
             optimizer.step()
         if isinstance(wrapped, DistributedDataParallel):
             wrapped.zero_grad_buffers()
-        else:
-            optimizer.zero_grad(set_to_none=False)
+        elif zero:
+            # set_to_none: AccumulateGrad steals the produced grad tensor —
+            # no per-param fill and no per-param add kernels
+            optimizer.zero_grad(set_to_none=True)
 
     graph = None
     if ns.graph and use_cuda and world == 1:
@@ -144,14 +146,18 @@ def main():
                 opt_step()
         torch.cuda.current_stream().wait_stream(s)
         graph = torch.cuda.CUDAGraph()
-        optimizer.zero_grad(set_to_none=False)
+        # capture with grads None: AccumulateGrad *assigns* inside the graph,
+        # so every replay overwrites the same grad blocks — no zero_grad and
+        # no accumulate-adds, ever (valid because each param has exactly one
+        # grad contribution in this model)
+        optimizer.zero_grad(set_to_none=True)
         with torch.cuda.graph(graph):
             static_loss = fwd_bwd()
 
         def step():
             reseed_dropout()
             graph.replay()
-            opt_step()
+            opt_step(zero=False)
             return static_loss
     else:
         def step():

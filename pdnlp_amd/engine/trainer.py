@@ -128,7 +128,7 @@ class Trainer:
         if isinstance(self.model, DistributedDataParallel):
             self.model.zero_grad_buffers()
         else:
-            self.optimizer.zero_grad(set_to_none=False)
+            self.optimizer.zero_grad(set_to_none=True)
 
     # ------------------------------------------------------------------
     def train(self, train_loader, dev_loader=None, train_sampler=None):
