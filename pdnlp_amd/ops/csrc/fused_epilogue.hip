@@ -3,11 +3,14 @@
 //
 //   out = LN(dropout(y + bias) + residual)
 //
-// One wavefront per row. Dropout mask from a counter-based hash RNG
-// (deterministic in (seed, element index), saved as u8 for backward —
-// SURVEY.md K16). The pre-LN sum is saved (bf16) so backward avoids
-// recomputing the dropout path. Five vendor-kernel passes in the reference
-// (bias add, dropout, residual add, LN) collapse to one HBM round trip.
+// One wavefront per row, four rows per 256-thread block, all global access
+// vectorized as short4 (8 B) — hipcc does not auto-vectorize scalar bf16
+// loads (guide §6 rule 2: scalar 2B loads halve effective HBM bandwidth).
+// Dropout mask from a counter-based hash RNG (deterministic in
+// (seed, element index), saved packed as u8x4 for backward — SURVEY.md K16).
+// The pre-LN sum is saved (bf16) so backward avoids recomputing the dropout
+// path. Five vendor-kernel passes in the reference (bias add, dropout,
+// residual add, LN stats, LN affine) collapse to one HBM round trip.
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
@@ -16,47 +19,71 @@
 
 namespace {
 
+template <typename T>
+__device__ __forceinline__ short4 ld4(const T* p) {
+  return *reinterpret_cast<const short4*>(p);
+}
+template <typename T>
+__device__ __forceinline__ void st4(T* p, short4 v) {
+  *reinterpret_cast<short4*>(p) = v;
+}
+template <typename T>
+__device__ __forceinline__ float elem(const short4& v, int j) {
+  return to_f32<T>(reinterpret_cast<const T*>(&v)[j]);
+}
+template <typename T>
+__device__ __forceinline__ void set_elem(short4& v, int j, float f) {
+  reinterpret_cast<T*>(&v)[j] = from_f32<T>(f);
+}
+
 // seed is read from DEVICE memory (seed_base) + a per-call-site salt so the
 // dropout mask changes across hipGraph replays (the host updates *seed_base
 // between replays; a kernel-arg seed would be frozen into the graph).
 template <typename T, bool DROP>
-__global__ void bdrl_fwd_kernel(const T* __restrict__ y,
-                                const T* __restrict__ bias,
-                                const T* __restrict__ res,
-                                const T* __restrict__ lnw,
-                                const T* __restrict__ lnb,
-                                T* __restrict__ out, T* __restrict__ xsum,
-                                unsigned char* __restrict__ mask_out,
-                                float* __restrict__ mean_out,
-                                float* __restrict__ rstd_out, int H, float p,
-                                float eps,
-                                const unsigned long long* __restrict__ seed_base,
-                                unsigned long long salt) {
+__global__ __launch_bounds__(256)
+void bdrl_fwd_kernel(const T* __restrict__ y, const T* __restrict__ bias,
+                     const T* __restrict__ res, const T* __restrict__ lnw,
+                     const T* __restrict__ lnb, T* __restrict__ out,
+                     T* __restrict__ xsum,
+                     unsigned char* __restrict__ mask_out,
+                     float* __restrict__ mean_out,
+                     float* __restrict__ rstd_out, int H, float p, float eps,
+                     const unsigned long long* __restrict__ seed_base,
+                     unsigned long long salt, long R) {
+  const long row = (long)blockIdx.x * 4 + (threadIdx.x >> 6);
+  if (row >= R) return;
   const unsigned long long seed = (DROP ? *seed_base : 0ull) + salt;
-  const long row = blockIdx.x;
   const int lane = threadIdx.x & (WAVE - 1);
   const T* yr = y + row * H;
   const T* rr = res + row * H;
   T* xr = xsum + row * H;
   T* outr = out + row * H;
-  const float keep = 1.f - p;
-  const float inv_keep = 1.f / keep;
+  const float inv_keep = 1.f / (1.f - p);
 
   float sum = 0.f, sumsq = 0.f;
-  for (int c = lane; c < H; c += WAVE) {
-    float h = to_f32<T>(yr[c]) + to_f32<T>(bias[c]);
-    if (DROP) {
-      const unsigned int r = hash_rng(seed, row * (unsigned long long)H + c);
-      const bool live = (r * 2.3283064365386963e-10f) >= p;
-      mask_out[row * H + c] = live;
-      h = live ? h * inv_keep : 0.f;
+  for (int c = lane * 4; c < H; c += WAVE * 4) {
+    const short4 yv = ld4(yr + c), bv = ld4(bias + c), rv = ld4(rr + c);
+    short4 xv;
+    uchar4 mv;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      float h = elem<T>(yv, j) + elem<T>(bv, j);
+      if (DROP) {
+        const unsigned int r =
+            hash_rng(seed, row * (unsigned long long)H + c + j);
+        const bool live = (r * 2.3283064365386963e-10f) >= p;
+        reinterpret_cast<unsigned char*>(&mv)[j] = live;
+        h = live ? h * inv_keep : 0.f;
+      }
+      h += elem<T>(rv, j);
+      set_elem<T>(xv, j, h);
+      // stats from the rounded value so they match the saved xsum exactly
+      const float hs = elem<T>(xv, j);
+      sum += hs;
+      sumsq += hs * hs;
     }
-    h += to_f32<T>(rr[c]);
-    xr[c] = from_f32<T>(h);
-    // recompute from the rounded value so stats match saved xsum exactly
-    const float hs = to_f32<T>(xr[c]);
-    sum += hs;
-    sumsq += hs * hs;
+    st4(xr + c, xv);
+    if (DROP) *reinterpret_cast<uchar4*>(mask_out + row * H + c) = mv;
   }
   sum = wave_sum(sum);
   sumsq = wave_sum(sumsq);
@@ -66,22 +93,28 @@ __global__ void bdrl_fwd_kernel(const T* __restrict__ y,
     mean_out[row] = mean;
     rstd_out[row] = rstd;
   }
-  for (int c = lane; c < H; c += WAVE) {
-    const float xh = (to_f32<T>(xr[c]) - mean) * rstd;
-    outr[c] = from_f32<T>(xh * to_f32<T>(lnw[c]) + to_f32<T>(lnb[c]));
+  for (int c = lane * 4; c < H; c += WAVE * 4) {
+    const short4 xv = ld4(xr + c), wv = ld4(lnw + c), bv = ld4(lnb + c);
+    short4 ov;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const float xh = (elem<T>(xv, j) - mean) * rstd;
+      set_elem<T>(ov, j, xh * elem<T>(wv, j) + elem<T>(bv, j));
+    }
+    st4(outr + c, ov);
   }
 }
 
 template <typename T, bool DROP>
-__global__ void bdrl_bwd_dx_kernel(const T* __restrict__ dout,
-                                   const T* __restrict__ xsum,
-                                   const unsigned char* __restrict__ mask,
-                                   const T* __restrict__ lnw,
-                                   const float* __restrict__ mean,
-                                   const float* __restrict__ rstd,
-                                   T* __restrict__ dy, T* __restrict__ dres,
-                                   int H, float p) {
-  const long row = blockIdx.x;
+__global__ __launch_bounds__(256)
+void bdrl_bwd_dx_kernel(const T* __restrict__ dout, const T* __restrict__ xsum,
+                        const unsigned char* __restrict__ mask,
+                        const T* __restrict__ lnw,
+                        const float* __restrict__ mean,
+                        const float* __restrict__ rstd, T* __restrict__ dy,
+                        T* __restrict__ dres, int H, float p, long R) {
+  const long row = (long)blockIdx.x * 4 + (threadIdx.x >> 6);
+  if (row >= R) return;
   const int lane = threadIdx.x & (WAVE - 1);
   const T* dor = dout + row * H;
   const T* xr = xsum + row * H;
@@ -91,53 +124,76 @@ __global__ void bdrl_bwd_dx_kernel(const T* __restrict__ dout,
   const float inv_keep = 1.f / (1.f - p);
 
   float s1 = 0.f, s2 = 0.f;
-  for (int c = lane; c < H; c += WAVE) {
-    const float dw = to_f32<T>(dor[c]) * to_f32<T>(lnw[c]);
-    const float xh = (to_f32<T>(xr[c]) - mu) * rs;
-    s1 += dw;
-    s2 += dw * xh;
+  for (int c = lane * 4; c < H; c += WAVE * 4) {
+    const short4 dv = ld4(dor + c), xv = ld4(xr + c), wv = ld4(lnw + c);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const float dw = elem<T>(dv, j) * elem<T>(wv, j);
+      const float xh = (elem<T>(xv, j) - mu) * rs;
+      s1 += dw;
+      s2 += dw * xh;
+    }
   }
   s1 = wave_sum(s1) / H;
   s2 = wave_sum(s2) / H;
-  for (int c = lane; c < H; c += WAVE) {
-    const float dw = to_f32<T>(dor[c]) * to_f32<T>(lnw[c]);
-    const float xh = (to_f32<T>(xr[c]) - mu) * rs;
-    const float dxs = rs * (dw - s1 - xh * s2);
-    drr[c] = from_f32<T>(dxs);
-    float g = dxs;
-    if (DROP) g = mask[row * H + c] ? g * inv_keep : 0.f;
-    dyr[c] = from_f32<T>(g);
+  for (int c = lane * 4; c < H; c += WAVE * 4) {
+    const short4 dv = ld4(dor + c), xv = ld4(xr + c), wv = ld4(lnw + c);
+    uchar4 mv;
+    if (DROP) mv = *reinterpret_cast<const uchar4*>(mask + row * H + c);
+    short4 dyv, drv;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const float dw = elem<T>(dv, j) * elem<T>(wv, j);
+      const float xh = (elem<T>(xv, j) - mu) * rs;
+      const float dxs = rs * (dw - s1 - xh * s2);
+      set_elem<T>(drv, j, dxs);
+      float g = dxs;
+      if (DROP)
+        g = reinterpret_cast<unsigned char*>(&mv)[j] ? g * inv_keep : 0.f;
+      set_elem<T>(dyv, j, g);
+    }
+    st4(drr + c, drv);
+    st4(dyr + c, dyv);
   }
 }
 
 // also folds the projection-bias gradient (column sum of dy, the
 // post-dropout grad produced by the dx kernel) into the same pass — saves a
-// separate torch reduce launch per call.
+// separate torch reduce launch per call. 4 columns per thread, short4 loads.
 template <typename T>
-__global__ void bdrl_bwd_dwdb_kernel(const T* __restrict__ dout,
-                                     const T* __restrict__ xsum,
-                                     const T* __restrict__ dy,
-                                     const float* __restrict__ mean,
-                                     const float* __restrict__ rstd,
-                                     float* __restrict__ dw32,
-                                     float* __restrict__ db32,
-                                     float* __restrict__ dbias32, long R,
-                                     int H, long rows_per_chunk) {
-  const int col = blockIdx.x * blockDim.x + threadIdx.x;
-  if (col >= H) return;
+__global__ __launch_bounds__(256)
+void bdrl_bwd_dwdb_kernel(const T* __restrict__ dout,
+                          const T* __restrict__ xsum,
+                          const T* __restrict__ dy,
+                          const float* __restrict__ mean,
+                          const float* __restrict__ rstd,
+                          float* __restrict__ dw32, float* __restrict__ db32,
+                          float* __restrict__ dbias32, long R, int H,
+                          long rows_per_chunk) {
+  const int c0 = (blockIdx.x * blockDim.x + threadIdx.x) * 4;
+  if (c0 >= H) return;
   const long r0 = blockIdx.y * rows_per_chunk;
   const long r1 = min(r0 + rows_per_chunk, R);
-  float dw = 0.f, db = 0.f, dbias = 0.f;
+  float dw[4] = {}, db[4] = {}, dbias[4] = {};
   for (long r = r0; r < r1; ++r) {
-    const float d = to_f32<T>(dout[r * H + col]);
-    const float xh = (to_f32<T>(xsum[r * H + col]) - mean[r]) * rstd[r];
-    dw += d * xh;
-    db += d;
-    dbias += to_f32<T>(dy[r * H + col]);
+    const short4 dv = ld4(dout + r * H + c0);
+    const short4 xv = ld4(xsum + r * H + c0);
+    const short4 yv = ld4(dy + r * H + c0);
+    const float mu = mean[r], rs = rstd[r];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const float d = elem<T>(dv, j);
+      dw[j] += d * ((elem<T>(xv, j) - mu) * rs);
+      db[j] += d;
+      dbias[j] += elem<T>(yv, j);
+    }
   }
-  atomicAdd(dw32 + col, dw);
-  atomicAdd(db32 + col, db);
-  atomicAdd(dbias32 + col, dbias);
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    atomicAdd(dw32 + c0 + j, dw[j]);
+    atomicAdd(db32 + c0 + j, db[j]);
+    atomicAdd(dbias32 + c0 + j, dbias[j]);
+  }
 }
 
 }  // namespace
@@ -148,6 +204,7 @@ std::vector<torch::Tensor> bias_dropout_residual_ln_fwd(
     long salt) {
   const int H = y.size(-1);
   const long R = y.numel() / H;
+  TORCH_CHECK(H % 4 == 0, "bdrl: hidden size must be a multiple of 4");
   auto out = torch::empty_like(y);
   auto xsum = torch::empty_like(y);
   const bool drop = p > 0.0;
@@ -161,10 +218,11 @@ std::vector<torch::Tensor> bias_dropout_residual_ln_fwd(
   auto mean = torch::empty({R}, y.options().dtype(torch::kFloat32));
   auto rstd = torch::empty({R}, y.options().dtype(torch::kFloat32));
   auto stream = at::hip::getCurrentHIPStream();
+  const long grid = (R + 3) / 4;
   DISPATCH_FLOAT_TYPES(y.scalar_type(), "bdrl_fwd", [&] {
     if (drop) {
-      hipLaunchKernelGGL((bdrl_fwd_kernel<scalar_t, true>), dim3(R),
-                         dim3(WAVE), 0, stream,
+      hipLaunchKernelGGL((bdrl_fwd_kernel<scalar_t, true>), dim3(grid),
+                         dim3(256), 0, stream,
                          (const scalar_t*)y.data_ptr(),
                          (const scalar_t*)bias.data_ptr(),
                          (const scalar_t*)res.data_ptr(),
@@ -176,10 +234,10 @@ std::vector<torch::Tensor> bias_dropout_residual_ln_fwd(
                          mean.data_ptr<float>(), rstd.data_ptr<float>(), H,
                          (float)p, (float)eps,
                          (const unsigned long long*)seed_buf.data_ptr(),
-                         (unsigned long long)salt);
+                         (unsigned long long)salt, R);
     } else {
-      hipLaunchKernelGGL((bdrl_fwd_kernel<scalar_t, false>), dim3(R),
-                         dim3(WAVE), 0, stream,
+      hipLaunchKernelGGL((bdrl_fwd_kernel<scalar_t, false>), dim3(grid),
+                         dim3(256), 0, stream,
                          (const scalar_t*)y.data_ptr(),
                          (const scalar_t*)bias.data_ptr(),
                          (const scalar_t*)res.data_ptr(),
@@ -188,7 +246,7 @@ std::vector<torch::Tensor> bias_dropout_residual_ln_fwd(
                          (scalar_t*)out.data_ptr(),
                          (scalar_t*)xsum.data_ptr(), nullptr,
                          mean.data_ptr<float>(), rstd.data_ptr<float>(), H,
-                         (float)p, (float)eps, nullptr, 0ull);
+                         (float)p, (float)eps, nullptr, 0ull, R);
     }
   });
   return {out, xsum, mask, mean, rstd};
@@ -199,38 +257,40 @@ std::vector<torch::Tensor> bias_dropout_residual_ln_bwd(
     torch::Tensor lnw, torch::Tensor mean, torch::Tensor rstd, double p) {
   const int H = xsum.size(-1);
   const long R = xsum.numel() / H;
+  TORCH_CHECK(H % 4 == 0, "bdrl: hidden size must be a multiple of 4");
   auto dy = torch::empty_like(xsum);
   auto dres = torch::empty_like(xsum);
-  auto dw32 = torch::zeros({H}, xsum.options().dtype(torch::kFloat32));
-  auto db32 = torch::zeros({H}, xsum.options().dtype(torch::kFloat32));
-  auto dbias32 = torch::zeros({H}, xsum.options().dtype(torch::kFloat32));
+  // one zero-filled [3, H] workspace instead of three separate fills
+  auto acc32 = torch::zeros({3, (long)H}, xsum.options().dtype(torch::kFloat32));
+  auto dw32 = acc32[0], db32 = acc32[1], dbias32 = acc32[2];
   auto stream = at::hip::getCurrentHIPStream();
   const bool drop = p > 0.0 && mask.numel() > 0;
   const long rows_per_chunk = 16;
   const long chunks = (R + rows_per_chunk - 1) / rows_per_chunk;
+  const long grid = (R + 3) / 4;
   DISPATCH_FLOAT_TYPES(xsum.scalar_type(), "bdrl_bwd", [&] {
     if (drop) {
-      hipLaunchKernelGGL((bdrl_bwd_dx_kernel<scalar_t, true>), dim3(R),
-                         dim3(WAVE), 0, stream,
+      hipLaunchKernelGGL((bdrl_bwd_dx_kernel<scalar_t, true>), dim3(grid),
+                         dim3(256), 0, stream,
                          (const scalar_t*)dout.data_ptr(),
                          (const scalar_t*)xsum.data_ptr(),
                          mask.data_ptr<unsigned char>(),
                          (const scalar_t*)lnw.data_ptr(),
                          mean.data_ptr<float>(), rstd.data_ptr<float>(),
                          (scalar_t*)dy.data_ptr(), (scalar_t*)dres.data_ptr(),
-                         H, (float)p);
+                         H, (float)p, R);
     } else {
-      hipLaunchKernelGGL((bdrl_bwd_dx_kernel<scalar_t, false>), dim3(R),
-                         dim3(WAVE), 0, stream,
+      hipLaunchKernelGGL((bdrl_bwd_dx_kernel<scalar_t, false>), dim3(grid),
+                         dim3(256), 0, stream,
                          (const scalar_t*)dout.data_ptr(),
                          (const scalar_t*)xsum.data_ptr(), nullptr,
                          (const scalar_t*)lnw.data_ptr(),
                          mean.data_ptr<float>(), rstd.data_ptr<float>(),
                          (scalar_t*)dy.data_ptr(), (scalar_t*)dres.data_ptr(),
-                         H, 0.f);
+                         H, 0.f, R);
     }
-    dim3 grid((H + 255) / 256, chunks);
-    hipLaunchKernelGGL((bdrl_bwd_dwdb_kernel<scalar_t>), grid, dim3(256), 0,
+    dim3 g2((H / 4 + 255) / 256, chunks);
+    hipLaunchKernelGGL((bdrl_bwd_dwdb_kernel<scalar_t>), g2, dim3(256), 0,
                        stream,
                        (const scalar_t*)dout.data_ptr(),
                        (const scalar_t*)xsum.data_ptr(),
@@ -239,6 +299,6 @@ std::vector<torch::Tensor> bias_dropout_residual_ln_bwd(
                        dw32.data_ptr<float>(), db32.data_ptr<float>(),
                        dbias32.data_ptr<float>(), R, H, rows_per_chunk);
   });
-  auto dt = lnw.scalar_type();
-  return {dy, dbias32.to(dt), dres, dw32.to(dt), db32.to(dt)};
+  auto accT = acc32.to(lnw.scalar_type());  // one cast kernel for all three
+  return {dy, accT[2], dres, accT[0], accT[1]};
 }

@@ -3,9 +3,10 @@
 // the bias/GELU/tanh epilogue fused into the C-write.
 //
 // Structure (the "step-3" ladder structure of the CDNA4 guide, §5):
-//  - 128x128 output tile, BK=64, 4 waves (256 threads), each wave a 64x64
-//    sub-tile = 4x4 fragments of v_mfma_f32_16x16x32_bf16 (f16 variant for
-//    fp16), fp32 accumulation in AGPRs.
+//  - templated BM x BN output tile, BK=64, 4 waves (256 threads); wave grid
+//    WMxWN, each wave an (BM/WM)x(BN/WN) sub-tile of
+//    v_mfma_f32_16x16x32_bf16 fragments (f16 variant for fp16), fp32
+//    accumulation in AGPRs.
 //  - global->LDS staging via __builtin_amdgcn_global_load_lds width 16
 //    (2 LDS buffers, stage tile t+1 while computing tile t, one
 //    vmcnt(0)+barrier per tile).
@@ -14,6 +15,11 @@
 //    and on the ds_read fragment address — ≤2-way instead of 8-way.
 //  - XCD-aware block swizzle (T1): contiguous tile chunks per XCD for L2
 //    affinity (bijective variant).
+//  - Tile shape picked per GEMM shape at launch: 128x128 when the grid
+//    fills the 256 CUs, 64x128 / 128x64 for skinny-N/M shapes where the
+//    128x128 grid would underfill the chip (e.g. the attention-output and
+//    FFN-down projections at N=768: 192 WGs -> 384 WGs).
+//    Override for sweeps: env PDNLP_GEMM_TILE=MxN (e.g. 64x128).
 //
 // A-fragment: lane l holds A[l&15][(l>>4)*8 + i], i=0..7 -> one ds_read_b128.
 // B-fragment: lane l holds W[n0 + (l&15)][k0 + (l>>4)*8 + i] — the same
@@ -26,6 +32,8 @@
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 
+#include <cstdlib>
+
 #include "common.h"
 
 namespace {
@@ -34,18 +42,18 @@ typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
 typedef _Float16 f16x8 __attribute__((ext_vector_type(8)));
 typedef float f32x4 __attribute__((ext_vector_type(4)));
 
-constexpr int BM = 128, BN = 128, BK = 64;
+constexpr int BK = 64;
 constexpr int NTHREADS = 256;
 
 __device__ __forceinline__ float gelu_f2(float x) {
   return 0.5f * x * (1.f + erff(x * 0.70710678118654752f));
 }
 
-// stage a BM x BK tile (rows of `src`, row stride `ld` elements) into LDS via
-// global_load_lds. Linear LDS image [128 rows][128 bytes]; the XOR swizzle is
+// stage a ROWS x BK tile (row stride `ld` elements) into LDS via
+// global_load_lds. Linear LDS image [ROWS][128 bytes]; the XOR swizzle is
 // pre-applied on the source byte offset. Each wave-instruction moves 8 rows
-// (8 lanes of 16 B per row); 4 waves x 4 calls cover 128 rows.
-template <typename T>
+// (8 lanes of 16 B per row); 4 waves x (ROWS/32) calls cover ROWS rows.
+template <typename T, int ROWS>
 __device__ __forceinline__ void stage_tile(const T* __restrict__ src, long ld,
                                            long row0, long max_row, long k0,
                                            char* lds) {
@@ -55,14 +63,14 @@ __device__ __forceinline__ void stage_tile(const T* __restrict__ src, long ld,
   const int piece = lane & 7;              // 16B piece within the 128B row
   const int kbyte = (piece * 16) ^ (sub_row << 4);  // source pre-swizzle
 #pragma unroll
-  for (int c = 0; c < 4; ++c) {
-    const int r = (wid * 4 + c) * 8 + sub_row;   // tile row 0..127
+  for (int c = 0; c < ROWS / 32; ++c) {
+    const int r = (wid * (ROWS / 32) + c) * 8 + sub_row;
     long gr = row0 + r;
     gr = gr < max_row ? gr : max_row - 1;        // clamp tail (stores guard)
     const char* gp = (const char*)(src + gr * ld + k0) + kbyte;
     // LDS dest operand is WAVE-UNIFORM (start of this call's 8-row chunk);
     // hardware writes lane l at dest + l*16 = row (l>>3), piece (l&7).
-    char* lp = lds + (long)(wid * 4 + c) * 8 * 128;
+    char* lp = lds + (long)(wid * (ROWS / 32) + c) * 8 * 128;
     __builtin_amdgcn_global_load_lds(
         (const __attribute__((address_space(1))) unsigned int*)gp,
         (__attribute__((address_space(3))) unsigned int*)lp, 16, 0, 0);
@@ -81,12 +89,19 @@ __device__ __forceinline__ V8 read_frag(const char* lds, int frag_row0,
 
 enum Act { ACT_NONE = 0, ACT_GELU = 1, ACT_TANH = 2 };
 
-template <typename T, typename V8, bool HAS_BIAS, int ACT, bool SAVE_PRE>
+template <typename T, typename V8, bool HAS_BIAS, int ACT, bool SAVE_PRE,
+          int BM, int BN>
 __global__ __launch_bounds__(NTHREADS)
 void gemm_nt_kernel(const T* __restrict__ A, const T* __restrict__ W,
                     const T* __restrict__ bias, T* __restrict__ C,
                     T* __restrict__ pre, long M, long N, long K,
                     int tiles_n, int nwg) {
+  // wave grid: 2x2 for square-ish tiles, 1x4 / 4x1 for skinny ones
+  constexpr int WM = (BM >= 128 || BN < 128) ? 2 : 1;
+  constexpr int WN = 4 / WM;
+  constexpr int TM = BM / WM, TN = BN / WN;     // per-wave sub-tile
+  constexpr int RM = TM / 16, RN = TN / 16;     // fragment repeats
+
   // XCD-aware bijective remap of the tile id (guide T1)
   int wg = blockIdx.x;
   {
@@ -98,16 +113,17 @@ void gemm_nt_kernel(const T* __restrict__ A, const T* __restrict__ W,
   const long tile_m = wg / tiles_n, tile_n = wg % tiles_n;
   const long m0 = tile_m * BM, n0 = tile_n * BN;
 
-  __shared__ __attribute__((aligned(16))) char lds[2][2][BM * 128];  // [buf][A/B]
+  __shared__ __attribute__((aligned(16))) char lds_a[2][BM * 128];
+  __shared__ __attribute__((aligned(16))) char lds_b[2][BN * 128];
 
   const int wid = threadIdx.x >> 6;
   const int lane = threadIdx.x & (WAVE - 1);
-  const int wr = (wid >> 1) * 64, wc = (wid & 1) * 64;  // wave sub-tile origin
+  const int wr = (wid / WN) * TM, wc = (wid % WN) * TN;
 
-  f32x4 acc[4][4] = {};
+  f32x4 acc[RM][RN] = {};
 
-  stage_tile<T>(A, K, m0, M, 0, lds[0][0]);
-  stage_tile<T>(W, K, n0, N, 0, lds[0][1]);
+  stage_tile<T, BM>(A, K, m0, M, 0, lds_a[0]);
+  stage_tile<T, BN>(W, K, n0, N, 0, lds_b[0]);
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
 
@@ -115,22 +131,22 @@ void gemm_nt_kernel(const T* __restrict__ A, const T* __restrict__ W,
   int cur = 0;
   for (int t = 0; t < ntiles; ++t) {
     if (t + 1 < ntiles) {
-      stage_tile<T>(A, K, m0, M, (long)(t + 1) * BK, lds[cur ^ 1][0]);
-      stage_tile<T>(W, K, n0, N, (long)(t + 1) * BK, lds[cur ^ 1][1]);
+      stage_tile<T, BM>(A, K, m0, M, (long)(t + 1) * BK, lds_a[cur ^ 1]);
+      stage_tile<T, BN>(W, K, n0, N, (long)(t + 1) * BK, lds_b[cur ^ 1]);
     }
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
-      V8 a_frag[4], b_frag[4];
+      V8 a_frag[RM], b_frag[RN];
 #pragma unroll
-      for (int i = 0; i < 4; ++i)
-        a_frag[i] = read_frag<V8>(lds[cur][0], wr + i * 16, ks);
+      for (int i = 0; i < RM; ++i)
+        a_frag[i] = read_frag<V8>(lds_a[cur], wr + i * 16, ks);
 #pragma unroll
-      for (int j = 0; j < 4; ++j)
-        b_frag[j] = read_frag<V8>(lds[cur][1], wc + j * 16, ks);
+      for (int j = 0; j < RN; ++j)
+        b_frag[j] = read_frag<V8>(lds_b[cur], wc + j * 16, ks);
 #pragma unroll
-      for (int i = 0; i < 4; ++i) {
+      for (int i = 0; i < RM; ++i) {
 #pragma unroll
-        for (int j = 0; j < 4; ++j) {
+        for (int j = 0; j < RN; ++j) {
           if constexpr (std::is_same<V8, bf16x8>::value) {
             acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                 a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
@@ -150,9 +166,9 @@ void gemm_nt_kernel(const T* __restrict__ A, const T* __restrict__ W,
   const int crow_off = (lane >> 4) * 4;
   const int ccol = lane & 15;
 #pragma unroll
-  for (int i = 0; i < 4; ++i) {
+  for (int i = 0; i < RM; ++i) {
 #pragma unroll
-    for (int j = 0; j < 4; ++j) {
+    for (int j = 0; j < RN; ++j) {
       const long n = n0 + wc + j * 16 + ccol;
       if (n >= N) continue;
       const float bv = HAS_BIAS ? to_f32<T>(bias[n]) : 0.f;
@@ -170,23 +186,55 @@ void gemm_nt_kernel(const T* __restrict__ A, const T* __restrict__ W,
   }
 }
 
+struct TileChoice { int bm, bn; };
+
+// pick the tile so the grid fills 256 CUs (>= ~2 WGs per CU preferred),
+// falling back to the highest-intensity 128x128 when everything fills.
+static TileChoice pick_tile(long M, long N) {
+  if (const char* env = std::getenv("PDNLP_GEMM_TILE")) {
+    int bm, bn;
+    if (std::sscanf(env, "%dx%d", &bm, &bn) == 2) return {bm, bn};
+  }
+  auto wgs = [&](int bm, int bn) {
+    return ((M + bm - 1) / bm) * ((N + bn - 1) / bn);
+  };
+  // swept on MI355X (profiles/r01 microbench): 128x128 only wins once the
+  // grid is deep (>=1024 WGs); small-K BERT shapes prefer smaller tiles
+  // with more workgroups; N<=768 prefers 64x64 (attnout 411 vs 261 TF).
+  if (wgs(128, 128) >= 1024) return {128, 128};
+  if (N <= 768) {
+    if (wgs(64, 64) >= 512) return {64, 64};
+    return {64, 128};
+  }
+  if (wgs(64, 128) >= 512) return {64, 128};
+  return {128, 128};
+}
+
 template <typename T, typename V8>
 void launch_gemm(const torch::Tensor& A, const torch::Tensor& W,
                  const torch::Tensor& bias, torch::Tensor& C,
                  torch::Tensor& pre, int act, bool has_bias, bool save_pre,
                  hipStream_t stream) {
   const long M = A.size(0), K = A.size(1), N = W.size(0);
-  const int tiles_m = (int)((M + BM - 1) / BM);
-  const int tiles_n = (int)((N + BN - 1) / BN);
+  const TileChoice tc = pick_tile(M, N);
+  const int tiles_m = (int)((M + tc.bm - 1) / tc.bm);
+  const int tiles_n = (int)((N + tc.bn - 1) / tc.bn);
   const int nwg = tiles_m * tiles_n;
   const T* bptr = has_bias ? (const T*)bias.data_ptr() : nullptr;
   T* pptr = save_pre ? (T*)pre.data_ptr() : nullptr;
 
+#define LAUNCH_T(HB, ACTV, SP, BMV, BNV)                                       \
+  hipLaunchKernelGGL((gemm_nt_kernel<T, V8, HB, ACTV, SP, BMV, BNV>),          \
+                     dim3(nwg), dim3(NTHREADS), 0, stream,                     \
+                     (const T*)A.data_ptr(), (const T*)W.data_ptr(), bptr,     \
+                     (T*)C.data_ptr(), pptr, M, N, K, tiles_n, nwg)
 #define LAUNCH(HB, ACTV, SP)                                                   \
-  hipLaunchKernelGGL((gemm_nt_kernel<T, V8, HB, ACTV, SP>), dim3(nwg),         \
-                     dim3(NTHREADS), 0, stream, (const T*)A.data_ptr(),        \
-                     (const T*)W.data_ptr(), bptr, (T*)C.data_ptr(), pptr, M,  \
-                     N, K, tiles_n, nwg)
+  do {                                                                         \
+    if (tc.bm == 64 && tc.bn == 128) LAUNCH_T(HB, ACTV, SP, 64, 128);          \
+    else if (tc.bm == 128 && tc.bn == 64) LAUNCH_T(HB, ACTV, SP, 128, 64);     \
+    else if (tc.bm == 64 && tc.bn == 64) LAUNCH_T(HB, ACTV, SP, 64, 64);       \
+    else LAUNCH_T(HB, ACTV, SP, 128, 128);                                     \
+  } while (0)
 
   if (act == ACT_NONE) {
     if (has_bias) LAUNCH(true, ACT_NONE, false);
@@ -199,6 +247,7 @@ void launch_gemm(const torch::Tensor& A, const torch::Tensor& W,
     else LAUNCH(false, ACT_TANH, true);
   }
 #undef LAUNCH
+#undef LAUNCH_T
 }
 
 }  // namespace

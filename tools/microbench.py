@@ -36,18 +36,34 @@ def gemm_bench():
         ("qkv_large", 8192, 3072, 1024),
         ("ffn_up_large", 8192, 4096, 1024),
     ]
+    import os
+    sweep = os.environ.get("SWEEP_TILES", "0") == "1"
     for name, M, N, K in shapes:
         A = torch.randn(M, K, device=DEV, dtype=torch.bfloat16)
         W = torch.randn(N, K, device=DEV, dtype=torch.bfloat16)
         b = torch.randn(N, device=DEV, dtype=torch.bfloat16)
+        flops = 2.0 * M * N * K
+        tiles = {}
+        if sweep:
+            ref = torch.matmul(A.float(), W.t().float())
+            for tile in ["128x128", "64x128", "128x64", "64x64"]:
+                os.environ["PDNLP_GEMM_TILE"] = tile
+                out = e.gemm_nt_fwd(A, W, b, "none")[0]
+                ok = torch.allclose(out.float(), ref + b.float(),
+                                    rtol=3e-2, atol=3e-2)
+                us = timeit(lambda: e.gemm_nt_fwd(A, W, b, "none"))
+                tiles[tile] = {"us": round(us, 2),
+                               "tflops": round(flops / us / 1e6, 1),
+                               "ok": bool(ok)}
+            del os.environ["PDNLP_GEMM_TILE"]
         us_ours = timeit(lambda: e.gemm_nt_fwd(A, W, b, "none"))
         us_blas = timeit(lambda: torch.matmul(A, W.t()))
-        flops = 2.0 * M * N * K
         print(json.dumps({
             "bench": "gemm_nt", "shape": name, "M": M, "N": N, "K": K,
             "ours_us": round(us_ours, 2), "rocblas_us": round(us_blas, 2),
             "ours_tflops": round(flops / us_ours / 1e6, 1),
-            "rocblas_tflops": round(flops / us_blas / 1e6, 1)}), flush=True)
+            "rocblas_tflops": round(flops / us_blas / 1e6, 1),
+            **({"tiles": tiles} if tiles else {})}), flush=True)
         # backward dgemm shapes (rocBLAS path): dX = dy@W, dW = dy^T@x
         dy = torch.randn(M, N, device=DEV, dtype=torch.bfloat16)
         us_dx = timeit(lambda: torch.matmul(dy, W))
