@@ -19,20 +19,25 @@
 
 namespace {
 
+// 4-element vector chunk: 8 B for bf16/fp16, 16 B for fp32
+template <typename T> struct V4 { using type = short4; };
+template <> struct V4<float> { using type = float4; };
+template <typename T> using v4_t = typename V4<T>::type;
+
 template <typename T>
-__device__ __forceinline__ short4 ld4(const T* p) {
-  return *reinterpret_cast<const short4*>(p);
+__device__ __forceinline__ v4_t<T> ld4(const T* p) {
+  return *reinterpret_cast<const v4_t<T>*>(p);
 }
 template <typename T>
-__device__ __forceinline__ void st4(T* p, short4 v) {
-  *reinterpret_cast<short4*>(p) = v;
+__device__ __forceinline__ void st4(T* p, v4_t<T> v) {
+  *reinterpret_cast<v4_t<T>*>(p) = v;
 }
 template <typename T>
-__device__ __forceinline__ float elem(const short4& v, int j) {
+__device__ __forceinline__ float elem(const v4_t<T>& v, int j) {
   return to_f32<T>(reinterpret_cast<const T*>(&v)[j]);
 }
 template <typename T>
-__device__ __forceinline__ void set_elem(short4& v, int j, float f) {
+__device__ __forceinline__ void set_elem(v4_t<T>& v, int j, float f) {
   reinterpret_cast<T*>(&v)[j] = from_f32<T>(f);
 }
 
@@ -62,8 +67,8 @@ void bdrl_fwd_kernel(const T* __restrict__ y, const T* __restrict__ bias,
 
   float sum = 0.f, sumsq = 0.f;
   for (int c = lane * 4; c < H; c += WAVE * 4) {
-    const short4 yv = ld4(yr + c), bv = ld4(bias + c), rv = ld4(rr + c);
-    short4 xv;
+    const v4_t<T> yv = ld4(yr + c), bv = ld4(bias + c), rv = ld4(rr + c);
+    v4_t<T> xv;
     uchar4 mv;
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
@@ -94,8 +99,8 @@ void bdrl_fwd_kernel(const T* __restrict__ y, const T* __restrict__ bias,
     rstd_out[row] = rstd;
   }
   for (int c = lane * 4; c < H; c += WAVE * 4) {
-    const short4 xv = ld4(xr + c), wv = ld4(lnw + c), bv = ld4(lnb + c);
-    short4 ov;
+    const v4_t<T> xv = ld4(xr + c), wv = ld4(lnw + c), bv = ld4(lnb + c);
+    v4_t<T> ov;
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
       const float xh = (elem<T>(xv, j) - mean) * rstd;
@@ -125,7 +130,7 @@ void bdrl_bwd_dx_kernel(const T* __restrict__ dout, const T* __restrict__ xsum,
 
   float s1 = 0.f, s2 = 0.f;
   for (int c = lane * 4; c < H; c += WAVE * 4) {
-    const short4 dv = ld4(dor + c), xv = ld4(xr + c), wv = ld4(lnw + c);
+    const v4_t<T> dv = ld4(dor + c), xv = ld4(xr + c), wv = ld4(lnw + c);
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
       const float dw = elem<T>(dv, j) * elem<T>(wv, j);
@@ -137,10 +142,10 @@ void bdrl_bwd_dx_kernel(const T* __restrict__ dout, const T* __restrict__ xsum,
   s1 = wave_sum(s1) / H;
   s2 = wave_sum(s2) / H;
   for (int c = lane * 4; c < H; c += WAVE * 4) {
-    const short4 dv = ld4(dor + c), xv = ld4(xr + c), wv = ld4(lnw + c);
+    const v4_t<T> dv = ld4(dor + c), xv = ld4(xr + c), wv = ld4(lnw + c);
     uchar4 mv;
     if (DROP) mv = *reinterpret_cast<const uchar4*>(mask + row * H + c);
-    short4 dyv, drv;
+    v4_t<T> dyv, drv;
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
       const float dw = elem<T>(dv, j) * elem<T>(wv, j);
@@ -176,9 +181,9 @@ void bdrl_bwd_dwdb_kernel(const T* __restrict__ dout,
   const long r1 = min(r0 + rows_per_chunk, R);
   float dw[4] = {}, db[4] = {}, dbias[4] = {};
   for (long r = r0; r < r1; ++r) {
-    const short4 dv = ld4(dout + r * H + c0);
-    const short4 xv = ld4(xsum + r * H + c0);
-    const short4 yv = ld4(dy + r * H + c0);
+    const v4_t<T> dv = ld4(dout + r * H + c0);
+    const v4_t<T> xv = ld4(xsum + r * H + c0);
+    const v4_t<T> yv = ld4(dy + r * H + c0);
     const float mu = mean[r], rs = rstd[r];
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
