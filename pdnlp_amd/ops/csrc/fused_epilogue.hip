@@ -117,87 +117,87 @@ void bdrl_bwd_dx_kernel(const T* __restrict__ dout, const T* __restrict__ xsum,
                         const T* __restrict__ lnw,
                         const float* __restrict__ mean,
                         const float* __restrict__ rstd, T* __restrict__ dy,
-                        T* __restrict__ dres, int H, float p, long R) {
-  const long row = (long)blockIdx.x * 4 + (threadIdx.x >> 6);
-  if (row >= R) return;
+                        T* __restrict__ dres, float* __restrict__ dw32,
+                        float* __restrict__ db32, float* __restrict__ dbias32,
+                        int H, float p, long R, int nblocks) {
+  // Persistent blocks (grid-stride over rows, wave per row) so the
+  // LN-weight/LN-bias/projection-bias column sums can be register-
+  // accumulated across all of this block's rows and flushed ONCE — the
+  // separate 19-MB second pass of the old dwdb kernel is gone.
+  // H <= 1024 (hidden size) => at most 4 column chunks per lane.
   const int lane = threadIdx.x & (WAVE - 1);
-  const T* dor = dout + row * H;
-  const T* xr = xsum + row * H;
-  T* dyr = dy + row * H;
-  T* drr = dres + row * H;
-  const float mu = mean[row], rs = rstd[row];
+  const int wid = threadIdx.x >> 6;
   const float inv_keep = 1.f / (1.f - p);
+  constexpr int MAXCH = 4;
+  const int nch = H / (WAVE * 4);
+  float adw[MAXCH][4] = {}, adb[MAXCH][4] = {}, adbias[MAXCH][4] = {};
 
-  float s1 = 0.f, s2 = 0.f;
-  for (int c = lane * 4; c < H; c += WAVE * 4) {
-    const v4_t<T> dv = ld4(dor + c), xv = ld4(xr + c), wv = ld4(lnw + c);
-#pragma unroll
-    for (int j = 0; j < 4; ++j) {
-      const float dw = elem<T>(dv, j) * elem<T>(wv, j);
-      const float xh = (elem<T>(xv, j) - mu) * rs;
-      s1 += dw;
-      s2 += dw * xh;
-    }
-  }
-  s1 = wave_sum(s1) / H;
-  s2 = wave_sum(s2) / H;
-  for (int c = lane * 4; c < H; c += WAVE * 4) {
-    const v4_t<T> dv = ld4(dor + c), xv = ld4(xr + c), wv = ld4(lnw + c);
-    uchar4 mv;
-    if (DROP) mv = *reinterpret_cast<const uchar4*>(mask + row * H + c);
-    v4_t<T> dyv, drv;
-#pragma unroll
-    for (int j = 0; j < 4; ++j) {
-      const float dw = elem<T>(dv, j) * elem<T>(wv, j);
-      const float xh = (elem<T>(xv, j) - mu) * rs;
-      const float dxs = rs * (dw - s1 - xh * s2);
-      set_elem<T>(drv, j, dxs);
-      float g = dxs;
-      if (DROP)
-        g = reinterpret_cast<unsigned char*>(&mv)[j] ? g * inv_keep : 0.f;
-      set_elem<T>(dyv, j, g);
-    }
-    st4(drr + c, drv);
-    st4(dyr + c, dyv);
-  }
-}
+  for (long row = (long)blockIdx.x * 4 + wid; row < R;
+       row += (long)nblocks * 4) {
+    const T* dor = dout + row * H;
+    const T* xr = xsum + row * H;
+    T* dyr = dy + row * H;
+    T* drr = dres + row * H;
+    const float mu = mean[row], rs = rstd[row];
 
-// also folds the projection-bias gradient (column sum of dy, the
-// post-dropout grad produced by the dx kernel) into the same pass — saves a
-// separate torch reduce launch per call. 4 columns per thread, short4 loads.
-template <typename T>
-__global__ __launch_bounds__(256)
-void bdrl_bwd_dwdb_kernel(const T* __restrict__ dout,
-                          const T* __restrict__ xsum,
-                          const T* __restrict__ dy,
-                          const float* __restrict__ mean,
-                          const float* __restrict__ rstd,
-                          float* __restrict__ dw32, float* __restrict__ db32,
-                          float* __restrict__ dbias32, long R, int H,
-                          long rows_per_chunk) {
-  const int c0 = (blockIdx.x * blockDim.x + threadIdx.x) * 4;
-  if (c0 >= H) return;
-  const long r0 = blockIdx.y * rows_per_chunk;
-  const long r1 = min(r0 + rows_per_chunk, R);
-  float dw[4] = {}, db[4] = {}, dbias[4] = {};
-  for (long r = r0; r < r1; ++r) {
-    const v4_t<T> dv = ld4(dout + r * H + c0);
-    const v4_t<T> xv = ld4(xsum + r * H + c0);
-    const v4_t<T> yv = ld4(dy + r * H + c0);
-    const float mu = mean[r], rs = rstd[r];
+    float s1 = 0.f, s2 = 0.f;
+    for (int c = lane * 4; c < H; c += WAVE * 4) {
+      const v4_t<T> dv = ld4(dor + c), xv = ld4(xr + c), wv = ld4(lnw + c);
 #pragma unroll
-    for (int j = 0; j < 4; ++j) {
-      const float d = elem<T>(dv, j);
-      dw[j] += d * ((elem<T>(xv, j) - mu) * rs);
-      db[j] += d;
-      dbias[j] += elem<T>(yv, j);
+      for (int j = 0; j < 4; ++j) {
+        const float dw = elem<T>(dv, j) * elem<T>(wv, j);
+        const float xh = (elem<T>(xv, j) - mu) * rs;
+        s1 += dw;
+        s2 += dw * xh;
+      }
+    }
+    s1 = wave_sum(s1) / H;
+    s2 = wave_sum(s2) / H;
+    int ch = 0;
+    for (int c = lane * 4; c < H; c += WAVE * 4, ++ch) {
+      const v4_t<T> dv = ld4(dor + c), xv = ld4(xr + c), wv = ld4(lnw + c);
+      uchar4 mv;
+      if (DROP) mv = *reinterpret_cast<const uchar4*>(mask + row * H + c);
+      v4_t<T> dyv, drv;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const float d = elem<T>(dv, j);
+        const float dw = d * elem<T>(wv, j);
+        const float xh = (elem<T>(xv, j) - mu) * rs;
+        const float dxs = rs * (dw - s1 - xh * s2);
+        set_elem<T>(drv, j, dxs);
+        float g = dxs;
+        if (DROP)
+          g = reinterpret_cast<unsigned char*>(&mv)[j] ? g * inv_keep : 0.f;
+        set_elem<T>(dyv, j, g);
+        adw[ch][j] += d * xh;
+        adb[ch][j] += d;
+        adbias[ch][j] += g;
+      }
+      st4(drr + c, drv);
+      st4(dyr + c, dyv);
     }
   }
+
+  // cross-wave reduce via LDS, then one global atomicAdd per column
+  __shared__ float lacc[3][1024];
+  for (int i = threadIdx.x; i < 3 * 1024; i += 256)
+    (&lacc[0][0])[i] = 0.f;
+  __syncthreads();
+  for (int ch = 0; ch < nch; ++ch) {
 #pragma unroll
-  for (int j = 0; j < 4; ++j) {
-    atomicAdd(dw32 + c0 + j, dw[j]);
-    atomicAdd(db32 + c0 + j, db[j]);
-    atomicAdd(dbias32 + c0 + j, dbias[j]);
+    for (int j = 0; j < 4; ++j) {
+      const int col = ch * (WAVE * 4) + lane * 4 + j;
+      atomicAdd(&lacc[0][col], adw[ch][j]);
+      atomicAdd(&lacc[1][col], adb[ch][j]);
+      atomicAdd(&lacc[2][col], adbias[ch][j]);
+    }
+  }
+  __syncthreads();
+  for (int col = threadIdx.x; col < H; col += 256) {
+    atomicAdd(dw32 + col, lacc[0][col]);
+    atomicAdd(db32 + col, lacc[1][col]);
+    atomicAdd(dbias32 + col, lacc[2][col]);
   }
 }
 
@@ -262,7 +262,8 @@ std::vector<torch::Tensor> bias_dropout_residual_ln_bwd(
     torch::Tensor lnw, torch::Tensor mean, torch::Tensor rstd, double p) {
   const int H = xsum.size(-1);
   const long R = xsum.numel() / H;
-  TORCH_CHECK(H % 4 == 0, "bdrl: hidden size must be a multiple of 4");
+  TORCH_CHECK(H % 256 == 0 && H <= 1024,
+              "bdrl: hidden size must be a multiple of 256 and <= 1024");
   auto dy = torch::empty_like(xsum);
   auto dres = torch::empty_like(xsum);
   // one zero-filled [3, H] workspace instead of three separate fills
@@ -270,12 +271,10 @@ std::vector<torch::Tensor> bias_dropout_residual_ln_bwd(
   auto dw32 = acc32[0], db32 = acc32[1], dbias32 = acc32[2];
   auto stream = at::hip::getCurrentHIPStream();
   const bool drop = p > 0.0 && mask.numel() > 0;
-  const long rows_per_chunk = 16;
-  const long chunks = (R + rows_per_chunk - 1) / rows_per_chunk;
-  const long grid = (R + 3) / 4;
+  const int nblocks = (int)std::min<long>((R + 3) / 4, 512);
   DISPATCH_FLOAT_TYPES(xsum.scalar_type(), "bdrl_bwd", [&] {
     if (drop) {
-      hipLaunchKernelGGL((bdrl_bwd_dx_kernel<scalar_t, true>), dim3(grid),
+      hipLaunchKernelGGL((bdrl_bwd_dx_kernel<scalar_t, true>), dim3(nblocks),
                          dim3(256), 0, stream,
                          (const scalar_t*)dout.data_ptr(),
                          (const scalar_t*)xsum.data_ptr(),
@@ -283,26 +282,19 @@ std::vector<torch::Tensor> bias_dropout_residual_ln_bwd(
                          (const scalar_t*)lnw.data_ptr(),
                          mean.data_ptr<float>(), rstd.data_ptr<float>(),
                          (scalar_t*)dy.data_ptr(), (scalar_t*)dres.data_ptr(),
-                         H, (float)p, R);
+                         dw32.data_ptr<float>(), db32.data_ptr<float>(),
+                         dbias32.data_ptr<float>(), H, (float)p, R, nblocks);
     } else {
-      hipLaunchKernelGGL((bdrl_bwd_dx_kernel<scalar_t, false>), dim3(grid),
+      hipLaunchKernelGGL((bdrl_bwd_dx_kernel<scalar_t, false>), dim3(nblocks),
                          dim3(256), 0, stream,
                          (const scalar_t*)dout.data_ptr(),
                          (const scalar_t*)xsum.data_ptr(), nullptr,
                          (const scalar_t*)lnw.data_ptr(),
                          mean.data_ptr<float>(), rstd.data_ptr<float>(),
                          (scalar_t*)dy.data_ptr(), (scalar_t*)dres.data_ptr(),
-                         H, 0.f, R);
+                         dw32.data_ptr<float>(), db32.data_ptr<float>(),
+                         dbias32.data_ptr<float>(), H, 0.f, R, nblocks);
     }
-    dim3 g2((H / 4 + 255) / 256, chunks);
-    hipLaunchKernelGGL((bdrl_bwd_dwdb_kernel<scalar_t>), g2, dim3(256), 0,
-                       stream,
-                       (const scalar_t*)dout.data_ptr(),
-                       (const scalar_t*)xsum.data_ptr(),
-                       (const scalar_t*)dy.data_ptr(),
-                       mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                       dw32.data_ptr<float>(), db32.data_ptr<float>(),
-                       dbias32.data_ptr<float>(), R, H, rows_per_chunk);
   });
   auto accT = acc32.to(lnw.scalar_type());  // one cast kernel for all three
   return {dy, accT[2], dres, accT[0], accT[1]};

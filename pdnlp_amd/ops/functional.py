@@ -104,13 +104,23 @@ class _LinearHipFn(torch.autograd.Function):
     def backward(ctx, dy):
         x2, w, pre = ctx.saved_tensors
         dy2 = dy.contiguous().view(-1, dy.shape[-1])
+        db = None
         if ctx.act == "gelu":
-            dy2 = ext().gelu_bwd(dy2, pre)
+            if ctx.has_bias and dy2.shape[-1] % 4 == 0:
+                # fused dgelu + bias-grad column sum (one pass)
+                dy2, db = ext().gelu_bwd_dbias(dy2, pre)
+            else:
+                dy2 = ext().gelu_bwd(dy2, pre)
         elif ctx.act == "tanh":
             dy2 = ext().tanh_bwd(dy2, pre)
         dx = dy2 @ w                      # rocBLAS NN
         dw = dy2.t() @ x2                 # rocBLAS TN
-        db = dy2.sum(0) if ctx.has_bias else None
+        if ctx.has_bias and db is None:
+            if dy2.shape[-1] % 4 == 0 and dy2.shape[0] >= 256 \
+                    and dy2.dtype != torch.float32:
+                db = ext().col_sum(dy2)
+            else:
+                db = dy2.sum(0)
         return dx.view(ctx.in_shape), dw, db, None
 
 
