@@ -153,8 +153,11 @@ void bdrl_bwd_dx_kernel(const T* __restrict__ dout, const T* __restrict__ xsum,
     }
     s1 = wave_sum(s1) / H;
     s2 = wave_sum(s2) / H;
-    int ch = 0;
-    for (int c = lane * 4; c < H; c += WAVE * 4, ++ch) {
+#pragma unroll
+    for (int ch = 0; ch < MAXCH; ++ch) {
+      // compile-time ch so the accumulator arrays stay in registers
+      const int c = lane * 4 + ch * (WAVE * 4);
+      if (c >= H) break;
       const v4_t<T> dv = ld4(dor + c), xv = ld4(xr + c), wv = ld4(lnw + c);
       uchar4 mv;
       if (DROP) mv = *reinterpret_cast<const uchar4*>(mask + row * H + c);

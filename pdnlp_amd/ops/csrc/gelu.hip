@@ -105,6 +105,7 @@ void gelu_bwd_dbias_kernel(const T* __restrict__ dy, const T* __restrict__ pre,
   const long r0 = blockIdx.y * rows_per_chunk;
   const long r1 = min(r0 + rows_per_chunk, R);
   float db[4] = {};
+#pragma unroll 4
   for (long r = r0; r < r1; ++r) {
     const short4 dv = *reinterpret_cast<const short4*>(dy + r * N + c0);
     const short4 pv = *reinterpret_cast<const short4*>(pre + r * N + c0);
@@ -135,6 +136,7 @@ void col_sum_kernel(const T* __restrict__ x, float* __restrict__ out32,
   const long r0 = blockIdx.y * rows_per_chunk;
   const long r1 = min(r0 + rows_per_chunk, R);
   float s[4] = {};
+#pragma unroll 4
   for (long r = r0; r < r1; ++r) {
     const short4 v = *reinterpret_cast<const short4*>(x + r * N + c0);
     const T* pv = reinterpret_cast<const T*>(&v);
@@ -222,7 +224,7 @@ std::vector<torch::Tensor> gelu_bwd_dbias(torch::Tensor dy, torch::Tensor pre) {
   auto dx = torch::empty_like(dy);
   auto db32 = torch::zeros({(long)N}, dy.options().dtype(torch::kFloat32));
   auto stream = at::hip::getCurrentHIPStream();
-  const long rows_per_chunk = 16;
+  const long rows_per_chunk = 64;
   const long chunks = (R + rows_per_chunk - 1) / rows_per_chunk;
   dim3 grid((N / 4 + 255) / 256, chunks);
   DISPATCH_FLOAT_TYPES(dy.scalar_type(), "gelu_bwd_dbias", [&] {
@@ -244,7 +246,7 @@ torch::Tensor col_sum(torch::Tensor x) {
               "col_sum: bf16/fp16, N % 4 == 0");
   auto out32 = torch::zeros({(long)N}, x.options().dtype(torch::kFloat32));
   auto stream = at::hip::getCurrentHIPStream();
-  const long rows_per_chunk = 16;
+  const long rows_per_chunk = 64;
   const long chunks = (R + rows_per_chunk - 1) / rows_per_chunk;
   dim3 grid((N / 4 + 255) / 256, chunks);
   DISPATCH_FLOAT_TYPES(x.scalar_type(), "col_sum", [&] {

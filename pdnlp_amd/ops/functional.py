@@ -349,7 +349,7 @@ def bias_dropout_residual_layernorm(
         training: bool = False, eps: float = 1e-12) -> torch.Tensor:
     """out = LN(dropout(y + bias) + residual) — the fused epilogue after the
     attention-output and FFN-down projections."""
-    if hip_enabled(y):
+    if hip_enabled(y) and y.shape[-1] % 256 == 0 and y.shape[-1] <= 1024:
         return _BiasDropResLNFn.apply(y, bias, residual, ln_w, ln_b, p,
                                       training, eps)
     h = y + bias
