@@ -42,6 +42,12 @@ def init_distributed(backend: Optional[str] = None,
         return local
 
     backend = backend or _default_backend()
+    # bind the device BEFORE creating the process group so RCCL communicators
+    # are built against the right HIP device (one process per GPU over xGMI)
+    if torch.cuda.is_available():
+        pre_local = int(os.environ.get("LOCAL_RANK",
+                                       rank if rank is not None else 0))
+        torch.cuda.set_device(pre_local % max(torch.cuda.device_count(), 1))
     kwargs = dict(backend=backend,
                   timeout=datetime.timedelta(seconds=timeout_sec))
     if init_method is not None:
