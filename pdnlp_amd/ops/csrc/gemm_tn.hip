@@ -1,0 +1,199 @@
+// Hand-written CDNA4 TN GEMM: C[N,K] = A[M,N]^T @ B[M,K] — the dW
+// weight-gradient GEMMs of the BERT backward (dW = dY^T X).
+// hipBLASLt tops out at ~300-450 TF on these shapes (small output, long
+// contraction; verified with PyTorch TunableOp exhaustive search), because
+// the M=4096 contraction has to stream through a small C tile.
+//
+// MI355X-native design:
+//  - 64x64 output tile, contraction chunked by 64, 4 waves (2x2) each
+//    owning a 32x32 sub-tile -> 432-576 workgroups on the BERT shapes
+//    (fills the 256 CUs without split-K).
+//  - Both operands are consumed TRANSPOSED (rows of dY^T / X^T). Instead of
+//    per-lane b16 gathers this uses gfx950's ds_read_b64_tr_b16 hardware
+//    transpose-read (guide T10): each 4-lane cluster reads 4 rows of a 4x4
+//    bf16 block (8-B aligned) and receives the block column-major.
+//  - The LDS image for a [32 m][16 col] subtile is the tr-read layout
+//    addr(m, c) = c + (m&3)*16 + (m>>3)*64 + ((m>>2)&1)*256 (elements),
+//    built directly by global_load_lds: dest element run l*8..l*8+7 decodes
+//    to (m_rel = ((l>>3)&3)*8 + (l>>5)*4 + ((l>>1)&3), c0 = (l&1)*8), i.e.
+//    each lane fetches 16 contiguous bytes of one global row — coalesced.
+//  - fp32 accumulation, bf16/fp16 C write, double-buffered staging.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "common.h"
+
+namespace {
+
+typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
+typedef _Float16 f16x8 __attribute__((ext_vector_type(8)));
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+typedef unsigned int uint2v __attribute__((ext_vector_type(2)));
+
+constexpr int NTHREADS = 256;
+constexpr int BT = 64;   // output tile is BT x BT
+constexpr int BC = 64;   // contraction chunk
+
+// stage a [BC m][BT cols] tile of `src` (row stride ld) into the tr-read
+// image. 8 subtiles of 512 elements; one glds wave-instruction fills one
+// subtile; 4 waves x 2 calls cover the tile. Rows clamped to max_m.
+template <typename T>
+__device__ __forceinline__ void stage_tr(const T* __restrict__ src, long ld,
+                                         long m0, long max_m, long col0,
+                                         char* lds) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x >> 6;
+  // lane -> (m_rel within 32, c half) inside one subtile
+  const int m_rel = ((lane >> 3) & 3) * 8 + (lane >> 5) * 4 + ((lane >> 1) & 3);
+  const int c0 = (lane & 1) * 8;
+#pragma unroll
+  for (int s = 0; s < 2; ++s) {
+    const int sub = wid * 2 + s;          // subtile 0..7
+    const int s_m = sub >> 2;             // m half (0: m 0..31, 1: 32..63)
+    const int s_c = sub & 3;              // 16-col group
+    long gm = m0 + s_m * 32 + m_rel;
+    gm = gm < max_m ? gm : max_m - 1;
+    const char* gp = (const char*)(src + gm * ld + col0 + s_c * 16 + c0);
+    char* lp = lds + sub * 1024;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)gp,
+        (__attribute__((address_space(3))) unsigned int*)lp, 16, 0, 0);
+  }
+}
+
+// transposed fragment via 2x ds_read_b64_tr_b16: lane l gets
+// data[ent0 + (l&15)][m0_sub + (l>>4)*8 + i], i = 0..7.
+// NOTE: caller must s_waitcnt lgkmcnt(0) before using the results (the
+// compiler cannot count asm ds ops).
+template <typename V8>
+__device__ __forceinline__ V8 frag_tr(const char* lds, int ent0, int m_sub) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int sub = (m_sub >> 5) * 4 + (ent0 >> 4);
+  const unsigned int base =
+      (unsigned int)(unsigned long)lds + sub * 1024 +
+      (lane & 3) * 32 + ((lane & 15) >> 2) * 8 + (lane >> 4) * 128;
+  uint2v lo, hi;
+  asm volatile("ds_read_b64_tr_b16 %0, %2\n\t"
+               "ds_read_b64_tr_b16 %1, %2 offset:512"
+               : "=v"(lo), "=v"(hi)
+               : "v"(base));
+  V8 out;
+  reinterpret_cast<uint2v*>(&out)[0] = lo;
+  reinterpret_cast<uint2v*>(&out)[1] = hi;
+  return out;
+}
+
+template <typename T, typename V8>
+__global__ __launch_bounds__(NTHREADS)
+void gemm_tn_kernel(const T* __restrict__ A, const T* __restrict__ B,
+                    T* __restrict__ C, long M, long N, long K,
+                    int tiles_k, int nwg) {
+  // XCD-aware bijective remap (guide T1)
+  int wg = blockIdx.x;
+  {
+    const int nxcd = 8;
+    const int q = nwg / nxcd, r = nwg % nxcd;
+    const int xcd = wg % nxcd, idx = wg / nxcd;
+    wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  }
+  const long n0 = (wg / tiles_k) * BT, k0 = (wg % tiles_k) * BT;
+
+  __shared__ __attribute__((aligned(16))) char lds_a[2][BC * BT * 2];
+  __shared__ __attribute__((aligned(16))) char lds_b[2][BC * BT * 2];
+
+  const int wid = threadIdx.x >> 6;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wr = (wid >> 1) * 32, wc = (wid & 1) * 32;  // 32x32 per wave
+
+  f32x4 acc[2][2] = {};
+
+  stage_tr<T>(A, N, 0, M, n0, lds_a[0]);
+  stage_tr<T>(B, K, 0, M, k0, lds_b[0]);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+
+  const int nchunks = (int)(M / BC);
+  int cur = 0;
+  for (int t = 0; t < nchunks; ++t) {
+    if (t + 1 < nchunks) {
+      stage_tr<T>(A, N, (long)(t + 1) * BC, M, n0, lds_a[cur ^ 1]);
+      stage_tr<T>(B, K, (long)(t + 1) * BC, M, k0, lds_b[cur ^ 1]);
+    }
+#pragma unroll
+    for (int ms = 0; ms < 2; ++ms) {
+      V8 a_frag[2], b_frag[2];
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+        a_frag[i] = frag_tr<V8>(lds_a[cur], wr + i * 16, ms * 32);
+#pragma unroll
+      for (int j = 0; j < 2; ++j)
+        b_frag[j] = frag_tr<V8>(lds_b[cur], wc + j * 16, ms * 32);
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+#pragma unroll
+      for (int i = 0; i < 2; ++i) {
+#pragma unroll
+        for (int j = 0; j < 2; ++j) {
+          if constexpr (std::is_same<V8, bf16x8>::value) {
+            acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
+          } else {
+            acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_f16(
+                a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
+          }
+        }
+      }
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  const int crow_off = (lane >> 4) * 4;
+  const int ccol = lane & 15;
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      const long k = k0 + wc + j * 16 + ccol;
+      if (k >= K) continue;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const long n = n0 + wr + i * 16 + crow_off + r;
+        if (n >= N) continue;
+        C[n * K + k] = from_f32<T>(acc[i][j][r]);
+      }
+    }
+  }
+}
+
+}  // namespace
+
+// C = A^T @ B for row-major A [M, N], B [M, K]; returns C [N, K] in A's
+// dtype (fp32 accumulation). Requires M % 64 == 0, N % 16 == 0, K % 16 == 0.
+torch::Tensor gemm_tn(torch::Tensor A, torch::Tensor B) {
+  TORCH_CHECK(A.is_cuda() && A.is_contiguous() && B.is_contiguous());
+  TORCH_CHECK(A.dim() == 2 && B.dim() == 2 && A.size(0) == B.size(0));
+  const long M = A.size(0), N = A.size(1), K = B.size(1);
+  TORCH_CHECK(M % BC == 0, "gemm_tn: M must be a multiple of 64");
+  auto C = torch::empty({N, K}, A.options());
+  const int tiles_n = (int)((N + BT - 1) / BT);
+  const int tiles_k = (int)((K + BT - 1) / BT);
+  const int nwg = tiles_n * tiles_k;
+  auto stream = at::hip::getCurrentHIPStream();
+  if (A.scalar_type() == torch::kBFloat16) {
+    hipLaunchKernelGGL((gemm_tn_kernel<__hip_bfloat16, bf16x8>), dim3(nwg),
+                       dim3(NTHREADS), 0, stream,
+                       (const __hip_bfloat16*)A.data_ptr(),
+                       (const __hip_bfloat16*)B.data_ptr(),
+                       (__hip_bfloat16*)C.data_ptr(), M, N, K, tiles_k, nwg);
+  } else if (A.scalar_type() == torch::kHalf) {
+    hipLaunchKernelGGL((gemm_tn_kernel<__half, f16x8>), dim3(nwg),
+                       dim3(NTHREADS), 0, stream,
+                       (const __half*)A.data_ptr(), (const __half*)B.data_ptr(),
+                       (__half*)C.data_ptr(), M, N, K, tiles_k, nwg);
+  } else {
+    TORCH_CHECK(false, "gemm_tn: bf16/fp16 only");
+  }
+  return C;
+}

@@ -442,3 +442,15 @@ def test_gelu_bwd_dbias_and_col_sum():
     cs = e.col_sum(dy)
     torch.testing.assert_close(cs.float(), dy.float().sum(0),
                                rtol=2e-2, atol=2e-1)
+
+
+@pytest.mark.parametrize("M,N,K", [(4096, 2304, 768), (4096, 768, 3072),
+                                   (512, 768, 768)])
+def test_gemm_tn(M, N, K):
+    e = ext()
+    torch.manual_seed(7)
+    A = torch.randn(M, N, device=DEV, dtype=torch.bfloat16)
+    B = torch.randn(M, K, device=DEV, dtype=torch.bfloat16)
+    C = e.gemm_tn(A, B)
+    ref = A.float().t() @ B.float()
+    torch.testing.assert_close(C.float(), ref, rtol=3e-2, atol=3e-1)
