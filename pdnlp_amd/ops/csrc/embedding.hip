@@ -61,8 +61,13 @@ __global__ void emb_ln_bwd_scatter_kernel(
     const T* __restrict__ word, const T* __restrict__ pos,
     const T* __restrict__ type_, const T* __restrict__ w,
     const float* __restrict__ mean, const float* __restrict__ rstd,
-    float* __restrict__ dword32, float* __restrict__ dpos32,
-    float* __restrict__ dtype32, int H) {
+    float* __restrict__ dword32, T* __restrict__ dsum_ws, int H) {
+  // One wave per token row. The LN-input gradient dsum is the gradient of
+  // ALL THREE tables at this row's indices; only the word table gets
+  // scatter-atomics here (random ids => low contention). Position/type
+  // rows are shared by many tokens (token_type 0 is shared by EVERY row:
+  // a 4096-way atomic pileup measured at ~190 us) — those tables are
+  // reduced from the dsum workspace by the column-parallel kernels below.
   const int row = blockIdx.x;
   const int lane = threadIdx.x & (WAVE - 1);
   const T* dyr = dy + (long)row * H;
@@ -82,17 +87,43 @@ __global__ void emb_ln_bwd_scatter_kernel(
   s1 = wave_sum(s1) / H;
   s2 = wave_sum(s2) / H;
   float* dwr = dword32 + (long)ids[row] * H;
-  float* dpr = dpos32 + (long)pos_ids[row] * H;
-  float* dtr = dtype32 + (long)type_ids[row] * H;
   for (int c = lane; c < H; c += WAVE) {
     const float dyw = to_f32<T>(dyr[c]) * to_f32<T>(w[c]);
     const float x = to_f32<T>(wr[c]) + to_f32<T>(pr[c]) + to_f32<T>(tr[c]);
     const float xh = (x - mu) * rs;
     const float dsum = rs * (dyw - s1 - xh * s2);
     atomicAdd(dwr + c, dsum);
-    atomicAdd(dpr + c, dsum);
-    atomicAdd(dtr + c, dsum);
+    dsum_ws[(long)row * H + c] = from_f32<T>(dsum);
   }
+}
+
+// dtab[v][c] = sum over rows with index v of dsum_ws[row][c], for a SMALL
+// table (position S<=512 rows, token-type 2 rows). Column-chunked: thread
+// owns one column over a row chunk, register accumulator per table row is
+// impossible generically, so accumulate per (row-index) via atomics into
+// the table — but now each chunk contributes ONE atomic per touched index
+// per column instead of one per token row.
+template <typename T>
+__global__ void emb_ln_bwd_small_table_kernel(
+    const T* __restrict__ dsum_ws, const long* __restrict__ idx,
+    float* __restrict__ dtab32, int R, int H, int rows_per_chunk,
+    int tab_rows) {
+  const int col = blockIdx.x * blockDim.x + threadIdx.x;
+  if (col >= H) return;
+  const int r0 = blockIdx.y * rows_per_chunk;
+  const int r1 = min(r0 + rows_per_chunk, R);
+  long cur = idx[r0];
+  float acc = 0.f;
+  for (int r = r0; r < r1; ++r) {
+    const long v = idx[r];
+    if (v != cur) {
+      atomicAdd(dtab32 + cur * H + col, acc);
+      cur = v;
+      acc = 0.f;
+    }
+    acc += to_f32<T>(dsum_ws[(long)r * H + col]);
+  }
+  atomicAdd(dtab32 + cur * H + col, acc);
 }
 
 template <typename T>
@@ -163,6 +194,7 @@ std::vector<torch::Tensor> embedding_ln_bwd(
   auto dtype32 = torch::zeros({type_.size(0), H}, opts32);
   auto dlnw32 = torch::zeros({H}, opts32);
   auto dlnb32 = torch::zeros({H}, opts32);
+  auto dsum_ws = torch::empty({R, (long)H}, word.options());
   auto stream = at::hip::getCurrentHIPStream();
   const int rows_per_chunk = 16;
   const int chunks = (int)((R + rows_per_chunk - 1) / rows_per_chunk);
@@ -176,10 +208,20 @@ std::vector<torch::Tensor> embedding_ln_bwd(
                        (const scalar_t*)type_.data_ptr(),
                        (const scalar_t*)w.data_ptr(), mean.data_ptr<float>(),
                        rstd.data_ptr<float>(), dword32.data_ptr<float>(),
-                       dpos32.data_ptr<float>(), dtype32.data_ptr<float>(), H);
-    dim3 grid((H + 255) / 256, chunks);
-    hipLaunchKernelGGL((emb_ln_bwd_dwdb_kernel<scalar_t>), grid, dim3(256), 0,
-                       stream,
+                       (scalar_t*)dsum_ws.data_ptr(), H);
+    dim3 gridc((H + 255) / 256, chunks);
+    hipLaunchKernelGGL((emb_ln_bwd_small_table_kernel<scalar_t>), gridc,
+                       dim3(256), 0, stream,
+                       (const scalar_t*)dsum_ws.data_ptr(),
+                       pos_ids.data_ptr<long>(), dpos32.data_ptr<float>(),
+                       (int)R, H, rows_per_chunk, (int)pos.size(0));
+    hipLaunchKernelGGL((emb_ln_bwd_small_table_kernel<scalar_t>), gridc,
+                       dim3(256), 0, stream,
+                       (const scalar_t*)dsum_ws.data_ptr(),
+                       type_ids.data_ptr<long>(), dtype32.data_ptr<float>(),
+                       (int)R, H, rows_per_chunk, (int)type_.size(0));
+    hipLaunchKernelGGL((emb_ln_bwd_dwdb_kernel<scalar_t>), gridc, dim3(256),
+                       0, stream,
                        (const scalar_t*)dy.data_ptr(), ids.data_ptr<long>(),
                        type_ids.data_ptr<long>(), pos_ids.data_ptr<long>(),
                        (const scalar_t*)word.data_ptr(),

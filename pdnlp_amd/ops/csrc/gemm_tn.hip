@@ -22,6 +22,8 @@
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 
+#include <cstdlib>
+
 #include "common.h"
 
 namespace {
@@ -62,29 +64,53 @@ __device__ __forceinline__ void stage_tr(const T* __restrict__ src, long ld,
   }
 }
 
-// transposed fragment via 2x ds_read_b64_tr_b16: lane l gets
-// data[ent0 + (l&15)][m0_sub + (l>>4)*8 + i], i = 0..7.
-// NOTE: caller must s_waitcnt lgkmcnt(0) before using the results (the
-// compiler cannot count asm ds ops).
-template <typename V8>
-__device__ __forceinline__ V8 frag_tr(const char* lds, int ent0, int m_sub) {
+// per-lane base address of the transposed fragment at (ent0, m_sub):
+// lane l will receive data[ent0 + (l&15)][m_sub + (l>>4)*8 + i], i = 0..7.
+__device__ __forceinline__ unsigned int frag_tr_base(const char* lds, int ent0,
+                                                     int m_sub) {
   const int lane = threadIdx.x & (WAVE - 1);
   const int sub = (m_sub >> 5) * 4 + (ent0 >> 4);
-  const unsigned int base =
-      (unsigned int)(unsigned long)lds + sub * 1024 +
-      (lane & 3) * 32 + ((lane & 15) >> 2) * 8 + (lane >> 4) * 128;
-  uint2v lo, hi;
-  asm volatile("ds_read_b64_tr_b16 %0, %2\n\t"
-               "ds_read_b64_tr_b16 %1, %2 offset:512"
-               : "=v"(lo), "=v"(hi)
-               : "v"(base));
-  V8 out;
-  reinterpret_cast<uint2v*>(&out)[0] = lo;
-  reinterpret_cast<uint2v*>(&out)[1] = hi;
-  return out;
+  // lane-LINEAR 8-B addresses over each quarter's 128-B [4m][16c] block —
+  // the hardware transposes within the block (guide T10: "no internal lane
+  // offset"); wrong per-lane math here silently feeds scrambled operands.
+  return (unsigned int)(unsigned long)lds + sub * 1024 + (lane & 15) * 8 +
+         (lane >> 4) * 128;
 }
 
-template <typename T, typename V8>
+// 4 transposed fragments (2 A + 2 B) in ONE asm block: 8x
+// ds_read_b64_tr_b16 then s_waitcnt lgkmcnt(0) INSIDE the block — the
+// compiler cannot count asm ds ops, so the wait must come before the
+// outputs escape the block.
+template <typename V8>
+__device__ __forceinline__ void frag_tr4(unsigned int a0, unsigned int a1,
+                                         unsigned int b0, unsigned int b1,
+                                         V8* af, V8* bf) {
+  uint2v r0, r1, r2, r3, r4, r5, r6, r7;
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %8\n\t"
+      "ds_read_b64_tr_b16 %1, %8 offset:512\n\t"
+      "ds_read_b64_tr_b16 %2, %9\n\t"
+      "ds_read_b64_tr_b16 %3, %9 offset:512\n\t"
+      "ds_read_b64_tr_b16 %4, %10\n\t"
+      "ds_read_b64_tr_b16 %5, %10 offset:512\n\t"
+      "ds_read_b64_tr_b16 %6, %11\n\t"
+      "ds_read_b64_tr_b16 %7, %11 offset:512\n\t"
+      "s_waitcnt lgkmcnt(0)"
+      : "=&v"(r0), "=&v"(r1), "=&v"(r2), "=&v"(r3), "=&v"(r4), "=&v"(r5),
+        "=&v"(r6), "=&v"(r7)
+      : "v"(a0), "v"(a1), "v"(b0), "v"(b1)
+      : "memory");
+  reinterpret_cast<uint2v*>(&af[0])[0] = r0;
+  reinterpret_cast<uint2v*>(&af[0])[1] = r1;
+  reinterpret_cast<uint2v*>(&af[1])[0] = r2;
+  reinterpret_cast<uint2v*>(&af[1])[1] = r3;
+  reinterpret_cast<uint2v*>(&bf[0])[0] = r4;
+  reinterpret_cast<uint2v*>(&bf[0])[1] = r5;
+  reinterpret_cast<uint2v*>(&bf[1])[0] = r6;
+  reinterpret_cast<uint2v*>(&bf[1])[1] = r7;
+}
+
+template <typename T, typename V8, bool PREFETCH>
 __global__ __launch_bounds__(NTHREADS)
 void gemm_tn_kernel(const T* __restrict__ A, const T* __restrict__ B,
                     T* __restrict__ C, long M, long N, long K,
@@ -116,20 +142,24 @@ void gemm_tn_kernel(const T* __restrict__ A, const T* __restrict__ B,
   const int nchunks = (int)(M / BC);
   int cur = 0;
   for (int t = 0; t < nchunks; ++t) {
-    if (t + 1 < nchunks) {
+    if (PREFETCH && t + 1 < nchunks) {
       stage_tr<T>(A, N, (long)(t + 1) * BC, M, n0, lds_a[cur ^ 1]);
       stage_tr<T>(B, K, (long)(t + 1) * BC, M, k0, lds_b[cur ^ 1]);
+    }
+    if (!PREFETCH && t > 0) {
+      stage_tr<T>(A, N, (long)t * BC, M, n0, lds_a[0]);
+      stage_tr<T>(B, K, (long)t * BC, M, k0, lds_b[0]);
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __syncthreads();
     }
 #pragma unroll
     for (int ms = 0; ms < 2; ++ms) {
       V8 a_frag[2], b_frag[2];
-#pragma unroll
-      for (int i = 0; i < 2; ++i)
-        a_frag[i] = frag_tr<V8>(lds_a[cur], wr + i * 16, ms * 32);
-#pragma unroll
-      for (int j = 0; j < 2; ++j)
-        b_frag[j] = frag_tr<V8>(lds_b[cur], wc + j * 16, ms * 32);
-      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      frag_tr4<V8>(frag_tr_base(lds_a[cur], wr, ms * 32),
+                   frag_tr_base(lds_a[cur], wr + 16, ms * 32),
+                   frag_tr_base(lds_b[cur], wc, ms * 32),
+                   frag_tr_base(lds_b[cur], wc + 16, ms * 32),
+                   a_frag, b_frag);
 #pragma unroll
       for (int i = 0; i < 2; ++i) {
 #pragma unroll
@@ -146,7 +176,7 @@ void gemm_tn_kernel(const T* __restrict__ A, const T* __restrict__ B,
     }
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __syncthreads();
-    cur ^= 1;
+    if (PREFETCH) cur ^= 1;
   }
 
   const int crow_off = (lane >> 4) * 4;
@@ -181,14 +211,22 @@ torch::Tensor gemm_tn(torch::Tensor A, torch::Tensor B) {
   const int tiles_k = (int)((K + BT - 1) / BT);
   const int nwg = tiles_n * tiles_k;
   auto stream = at::hip::getCurrentHIPStream();
+  const bool serial = getenv("PDNLP_TN_SERIAL") != nullptr;
   if (A.scalar_type() == torch::kBFloat16) {
-    hipLaunchKernelGGL((gemm_tn_kernel<__hip_bfloat16, bf16x8>), dim3(nwg),
-                       dim3(NTHREADS), 0, stream,
-                       (const __hip_bfloat16*)A.data_ptr(),
-                       (const __hip_bfloat16*)B.data_ptr(),
-                       (__hip_bfloat16*)C.data_ptr(), M, N, K, tiles_k, nwg);
+    if (serial)
+      hipLaunchKernelGGL((gemm_tn_kernel<__hip_bfloat16, bf16x8, false>),
+                         dim3(nwg), dim3(NTHREADS), 0, stream,
+                         (const __hip_bfloat16*)A.data_ptr(),
+                         (const __hip_bfloat16*)B.data_ptr(),
+                         (__hip_bfloat16*)C.data_ptr(), M, N, K, tiles_k, nwg);
+    else
+      hipLaunchKernelGGL((gemm_tn_kernel<__hip_bfloat16, bf16x8, true>),
+                         dim3(nwg), dim3(NTHREADS), 0, stream,
+                         (const __hip_bfloat16*)A.data_ptr(),
+                         (const __hip_bfloat16*)B.data_ptr(),
+                         (__hip_bfloat16*)C.data_ptr(), M, N, K, tiles_k, nwg);
   } else if (A.scalar_type() == torch::kHalf) {
-    hipLaunchKernelGGL((gemm_tn_kernel<__half, f16x8>), dim3(nwg),
+    hipLaunchKernelGGL((gemm_tn_kernel<__half, f16x8, true>), dim3(nwg),
                        dim3(NTHREADS), 0, stream,
                        (const __half*)A.data_ptr(), (const __half*)B.data_ptr(),
                        (__half*)C.data_ptr(), M, N, K, tiles_k, nwg);
