@@ -110,7 +110,7 @@ void bdrl_fwd_kernel(const T* __restrict__ y, const T* __restrict__ bias,
   }
 }
 
-template <typename T, bool DROP>
+template <typename T, bool DROP, int NCH>
 __global__ __launch_bounds__(256)
 void bdrl_bwd_dx_kernel(const T* __restrict__ dout, const T* __restrict__ xsum,
                         const unsigned char* __restrict__ mask,
@@ -128,9 +128,9 @@ void bdrl_bwd_dx_kernel(const T* __restrict__ dout, const T* __restrict__ xsum,
   const int lane = threadIdx.x & (WAVE - 1);
   const int wid = threadIdx.x >> 6;
   const float inv_keep = 1.f / (1.f - p);
-  constexpr int MAXCH = 4;
-  const int nch = H / (WAVE * 4);
-  float adw[MAXCH][4] = {}, adb[MAXCH][4] = {}, adbias[MAXCH][4] = {};
+  // NCH is compile-time (H/256 = 3 or 4) so the accumulators stay in
+  // registers — a runtime chunk index spills them to scratch (measured 4x)
+  float adw[NCH][4] = {}, adb[NCH][4] = {}, adbias[NCH][4] = {};
 
   for (long row = (long)blockIdx.x * 4 + wid; row < R;
        row += (long)nblocks * 4) {
@@ -154,10 +154,8 @@ void bdrl_bwd_dx_kernel(const T* __restrict__ dout, const T* __restrict__ xsum,
     s1 = wave_sum(s1) / H;
     s2 = wave_sum(s2) / H;
 #pragma unroll
-    for (int ch = 0; ch < MAXCH; ++ch) {
-      // compile-time ch so the accumulator arrays stay in registers
+    for (int ch = 0; ch < NCH; ++ch) {
       const int c = lane * 4 + ch * (WAVE * 4);
-      if (c >= H) break;
       const v4_t<T> dv = ld4(dor + c), xv = ld4(xr + c), wv = ld4(lnw + c);
       uchar4 mv;
       if (DROP) mv = *reinterpret_cast<const uchar4*>(mask + row * H + c);
@@ -187,7 +185,8 @@ void bdrl_bwd_dx_kernel(const T* __restrict__ dout, const T* __restrict__ xsum,
   for (int i = threadIdx.x; i < 3 * 1024; i += 256)
     (&lacc[0][0])[i] = 0.f;
   __syncthreads();
-  for (int ch = 0; ch < nch; ++ch) {
+#pragma unroll
+  for (int ch = 0; ch < NCH; ++ch) {
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
       const int col = ch * (WAVE * 4) + lane * 4 + j;
@@ -275,28 +274,30 @@ std::vector<torch::Tensor> bias_dropout_residual_ln_bwd(
   auto stream = at::hip::getCurrentHIPStream();
   const bool drop = p > 0.0 && mask.numel() > 0;
   const int nblocks = (int)std::min<long>((R + 3) / 4, 512);
+  TORCH_CHECK(H == 768 || H == 1024, "bdrl: hidden size must be 768 or 1024");
   DISPATCH_FLOAT_TYPES(xsum.scalar_type(), "bdrl_bwd", [&] {
+    auto launch = [&](auto drop_c, auto nch_c) {
+      hipLaunchKernelGGL(
+          (bdrl_bwd_dx_kernel<scalar_t, decltype(drop_c)::value,
+                              decltype(nch_c)::value>),
+          dim3(nblocks), dim3(256), 0, stream,
+          (const scalar_t*)dout.data_ptr(),
+          (const scalar_t*)xsum.data_ptr(),
+          drop ? mask.data_ptr<unsigned char>() : nullptr,
+          (const scalar_t*)lnw.data_ptr(), mean.data_ptr<float>(),
+          rstd.data_ptr<float>(), (scalar_t*)dy.data_ptr(),
+          (scalar_t*)dres.data_ptr(), dw32.data_ptr<float>(),
+          db32.data_ptr<float>(), dbias32.data_ptr<float>(), H,
+          drop ? (float)p : 0.f, R, nblocks);
+    };
+    using T3 = std::integral_constant<int, 3>;
+    using T4 = std::integral_constant<int, 4>;
     if (drop) {
-      hipLaunchKernelGGL((bdrl_bwd_dx_kernel<scalar_t, true>), dim3(nblocks),
-                         dim3(256), 0, stream,
-                         (const scalar_t*)dout.data_ptr(),
-                         (const scalar_t*)xsum.data_ptr(),
-                         mask.data_ptr<unsigned char>(),
-                         (const scalar_t*)lnw.data_ptr(),
-                         mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                         (scalar_t*)dy.data_ptr(), (scalar_t*)dres.data_ptr(),
-                         dw32.data_ptr<float>(), db32.data_ptr<float>(),
-                         dbias32.data_ptr<float>(), H, (float)p, R, nblocks);
+      if (H == 768) launch(std::true_type{}, T3{});
+      else launch(std::true_type{}, T4{});
     } else {
-      hipLaunchKernelGGL((bdrl_bwd_dx_kernel<scalar_t, false>), dim3(nblocks),
-                         dim3(256), 0, stream,
-                         (const scalar_t*)dout.data_ptr(),
-                         (const scalar_t*)xsum.data_ptr(), nullptr,
-                         (const scalar_t*)lnw.data_ptr(),
-                         mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                         (scalar_t*)dy.data_ptr(), (scalar_t*)dres.data_ptr(),
-                         dw32.data_ptr<float>(), db32.data_ptr<float>(),
-                         dbias32.data_ptr<float>(), H, 0.f, R, nblocks);
+      if (H == 768) launch(std::false_type{}, T3{});
+      else launch(std::false_type{}, T4{});
     }
   });
   auto accT = acc32.to(lnw.scalar_type());  // one cast kernel for all three

@@ -66,15 +66,38 @@ __global__ void bias_gelu_bwd_kernel(const T* __restrict__ dy,
   }
 }
 
-// dgelu on a saved pre-activation (GEMM fused-epilogue backward)
+// dgelu on a saved pre-activation (GEMM fused-epilogue backward) —
+// grid-stride, 8 elements / 16-B loads per thread iteration
 template <typename T>
 __global__ void gelu_bwd_kernel(const T* __restrict__ dy,
                                 const T* __restrict__ pre,
                                 T* __restrict__ dx, long total) {
-  const long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  const long stride = (long)gridDim.x * blockDim.x;
+  const long i0 = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  const long stride = (long)gridDim.x * blockDim.x * 8;
   for (long i = i0; i < total; i += stride) {
-    dx[i] = from_f32<T>(to_f32<T>(dy[i]) * gelu_grad_f(to_f32<T>(pre[i])));
+    if (sizeof(T) == 2 && i + 8 <= total) {
+      short4 d0 = *reinterpret_cast<const short4*>(dy + i);
+      short4 d1 = *reinterpret_cast<const short4*>(dy + i + 4);
+      short4 p0 = *reinterpret_cast<const short4*>(pre + i);
+      short4 p1 = *reinterpret_cast<const short4*>(pre + i + 4);
+      short4 o0, o1;
+      const T* pd0 = reinterpret_cast<const T*>(&d0);
+      const T* pd1 = reinterpret_cast<const T*>(&d1);
+      const T* pp0 = reinterpret_cast<const T*>(&p0);
+      const T* pp1 = reinterpret_cast<const T*>(&p1);
+      T* po0 = reinterpret_cast<T*>(&o0);
+      T* po1 = reinterpret_cast<T*>(&o1);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        po0[j] = from_f32<T>(to_f32<T>(pd0[j]) * gelu_grad_f(to_f32<T>(pp0[j])));
+        po1[j] = from_f32<T>(to_f32<T>(pd1[j]) * gelu_grad_f(to_f32<T>(pp1[j])));
+      }
+      *reinterpret_cast<short4*>(dx + i) = o0;
+      *reinterpret_cast<short4*>(dx + i + 4) = o1;
+    } else {
+      for (long k = i; k < min(i + 8, total); ++k)
+        dx[k] = from_f32<T>(to_f32<T>(dy[k]) * gelu_grad_f(to_f32<T>(pre[k])));
+    }
   }
 }
 
@@ -192,7 +215,7 @@ torch::Tensor gelu_bwd(torch::Tensor dy, torch::Tensor pre) {
   const long total = dy.numel();
   auto stream = at::hip::getCurrentHIPStream();
   DISPATCH_FLOAT_TYPES(dy.scalar_type(), "gelu_bwd", [&] {
-    hipLaunchKernelGGL((gelu_bwd_kernel<scalar_t>), dim3(grid_for(total)),
+    hipLaunchKernelGGL((gelu_bwd_kernel<scalar_t>), dim3(grid_for(total, 8)),
                        dim3(256), 0, stream,
                        (const scalar_t*)dy.data_ptr(),
                        (const scalar_t*)pre.data_ptr(),

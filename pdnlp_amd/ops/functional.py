@@ -106,11 +106,7 @@ class _LinearHipFn(torch.autograd.Function):
         dy2 = dy.contiguous().view(-1, dy.shape[-1])
         db = None
         if ctx.act == "gelu":
-            if ctx.has_bias and dy2.shape[-1] % 4 == 0:
-                # fused dgelu + bias-grad column sum (one pass)
-                dy2, db = ext().gelu_bwd_dbias(dy2, pre)
-            else:
-                dy2 = ext().gelu_bwd(dy2, pre)
+            dy2 = ext().gelu_bwd(dy2, pre)
         elif ctx.act == "tanh":
             dy2 = ext().tanh_bwd(dy2, pre)
         dx = dy2 @ w                      # rocBLAS NN
