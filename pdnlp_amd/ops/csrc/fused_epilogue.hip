@@ -110,97 +110,87 @@ void bdrl_fwd_kernel(const T* __restrict__ y, const T* __restrict__ bias,
   }
 }
 
-template <typename T, bool DROP, int NCH>
+template <typename T, bool DROP>
 __global__ __launch_bounds__(256)
 void bdrl_bwd_dx_kernel(const T* __restrict__ dout, const T* __restrict__ xsum,
                         const unsigned char* __restrict__ mask,
                         const T* __restrict__ lnw,
                         const float* __restrict__ mean,
                         const float* __restrict__ rstd, T* __restrict__ dy,
-                        T* __restrict__ dres, float* __restrict__ dw32,
-                        float* __restrict__ db32, float* __restrict__ dbias32,
-                        int H, float p, long R, int nblocks) {
-  // Persistent blocks (grid-stride over rows, wave per row) so the
-  // LN-weight/LN-bias/projection-bias column sums can be register-
-  // accumulated across all of this block's rows and flushed ONCE — the
-  // separate 19-MB second pass of the old dwdb kernel is gone.
-  // H <= 1024 (hidden size) => at most 4 column chunks per lane.
+                        T* __restrict__ dres, int H, float p, long R) {
+  const long row = (long)blockIdx.x * 4 + (threadIdx.x >> 6);
+  if (row >= R) return;
   const int lane = threadIdx.x & (WAVE - 1);
-  const int wid = threadIdx.x >> 6;
+  const T* dor = dout + row * H;
+  const T* xr = xsum + row * H;
+  T* dyr = dy + row * H;
+  T* drr = dres + row * H;
+  const float mu = mean[row], rs = rstd[row];
   const float inv_keep = 1.f / (1.f - p);
-  // NCH is compile-time (H/256 = 3 or 4) so the accumulators stay in
-  // registers — a runtime chunk index spills them to scratch (measured 4x)
-  float adw[NCH][4] = {}, adb[NCH][4] = {}, adbias[NCH][4] = {};
 
-  for (long row = (long)blockIdx.x * 4 + wid; row < R;
-       row += (long)nblocks * 4) {
-    const T* dor = dout + row * H;
-    const T* xr = xsum + row * H;
-    T* dyr = dy + row * H;
-    T* drr = dres + row * H;
-    const float mu = mean[row], rs = rstd[row];
-
-    float s1 = 0.f, s2 = 0.f;
-    for (int c = lane * 4; c < H; c += WAVE * 4) {
-      const v4_t<T> dv = ld4(dor + c), xv = ld4(xr + c), wv = ld4(lnw + c);
-#pragma unroll
-      for (int j = 0; j < 4; ++j) {
-        const float dw = elem<T>(dv, j) * elem<T>(wv, j);
-        const float xh = (elem<T>(xv, j) - mu) * rs;
-        s1 += dw;
-        s2 += dw * xh;
-      }
-    }
-    s1 = wave_sum(s1) / H;
-    s2 = wave_sum(s2) / H;
-#pragma unroll
-    for (int ch = 0; ch < NCH; ++ch) {
-      const int c = lane * 4 + ch * (WAVE * 4);
-      const v4_t<T> dv = ld4(dor + c), xv = ld4(xr + c), wv = ld4(lnw + c);
-      uchar4 mv;
-      if (DROP) mv = *reinterpret_cast<const uchar4*>(mask + row * H + c);
-      v4_t<T> dyv, drv;
-#pragma unroll
-      for (int j = 0; j < 4; ++j) {
-        const float d = elem<T>(dv, j);
-        const float dw = d * elem<T>(wv, j);
-        const float xh = (elem<T>(xv, j) - mu) * rs;
-        const float dxs = rs * (dw - s1 - xh * s2);
-        set_elem<T>(drv, j, dxs);
-        float g = dxs;
-        if (DROP)
-          g = reinterpret_cast<unsigned char*>(&mv)[j] ? g * inv_keep : 0.f;
-        set_elem<T>(dyv, j, g);
-        adw[ch][j] += d * xh;
-        adb[ch][j] += d;
-        adbias[ch][j] += g;
-      }
-      st4(drr + c, drv);
-      st4(dyr + c, dyv);
-    }
-  }
-
-  // cross-wave reduce via LDS, then one global atomicAdd per column
-  __shared__ float lacc[3][1024];
-  for (int i = threadIdx.x; i < 3 * 1024; i += 256)
-    (&lacc[0][0])[i] = 0.f;
-  __syncthreads();
-#pragma unroll
-  for (int ch = 0; ch < NCH; ++ch) {
+  float s1 = 0.f, s2 = 0.f;
+  for (int c = lane * 4; c < H; c += WAVE * 4) {
+    const v4_t<T> dv = ld4(dor + c), xv = ld4(xr + c), wv = ld4(lnw + c);
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
-      const int col = ch * (WAVE * 4) + lane * 4 + j;
-      atomicAdd(&lacc[0][col], adw[ch][j]);
-      atomicAdd(&lacc[1][col], adb[ch][j]);
-      atomicAdd(&lacc[2][col], adbias[ch][j]);
+      const float dw = elem<T>(dv, j) * elem<T>(wv, j);
+      const float xh = (elem<T>(xv, j) - mu) * rs;
+      s1 += dw;
+      s2 += dw * xh;
     }
   }
-  __syncthreads();
-  for (int col = threadIdx.x; col < H; col += 256) {
-    atomicAdd(dw32 + col, lacc[0][col]);
-    atomicAdd(db32 + col, lacc[1][col]);
-    atomicAdd(dbias32 + col, lacc[2][col]);
+  s1 = wave_sum(s1) / H;
+  s2 = wave_sum(s2) / H;
+  for (int c = lane * 4; c < H; c += WAVE * 4) {
+    const v4_t<T> dv = ld4(dor + c), xv = ld4(xr + c), wv = ld4(lnw + c);
+    uchar4 mv;
+    if (DROP) mv = *reinterpret_cast<const uchar4*>(mask + row * H + c);
+    v4_t<T> dyv, drv;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const float dw = elem<T>(dv, j) * elem<T>(wv, j);
+      const float xh = (elem<T>(xv, j) - mu) * rs;
+      const float dxs = rs * (dw - s1 - xh * s2);
+      set_elem<T>(drv, j, dxs);
+      float g = dxs;
+      if (DROP)
+        g = reinterpret_cast<unsigned char*>(&mv)[j] ? g * inv_keep : 0.f;
+      set_elem<T>(dyv, j, g);
+    }
+    st4(drr + c, drv);
+    st4(dyr + c, dyv);
   }
+}
+
+// LN-weight/LN-bias/projection-bias column sums in a second pass: one
+// column per thread (scalar 2-B loads stay fully coalesced across the 256
+// threads and the high thread count is what hides latency here — a 4-col
+// vector variant measured SLOWER), 32-row chunks, fp32 atomics.
+template <typename T>
+__global__ __launch_bounds__(256)
+void bdrl_bwd_dwdb_kernel(const T* __restrict__ dout,
+                          const T* __restrict__ xsum,
+                          const T* __restrict__ dy,
+                          const float* __restrict__ mean,
+                          const float* __restrict__ rstd,
+                          float* __restrict__ dw32, float* __restrict__ db32,
+                          float* __restrict__ dbias32, long R, int H,
+                          long rows_per_chunk) {
+  const int col = blockIdx.x * blockDim.x + threadIdx.x;
+  if (col >= H) return;
+  const long r0 = blockIdx.y * rows_per_chunk;
+  const long r1 = min(r0 + rows_per_chunk, R);
+  float dw = 0.f, db = 0.f, dbias = 0.f;
+  for (long r = r0; r < r1; ++r) {
+    const float d = to_f32<T>(dout[r * H + col]);
+    const float xh = (to_f32<T>(xsum[r * H + col]) - mean[r]) * rstd[r];
+    dw += d * xh;
+    db += d;
+    dbias += to_f32<T>(dy[r * H + col]);
+  }
+  atomicAdd(dw32 + col, dw);
+  atomicAdd(db32 + col, db);
+  atomicAdd(dbias32 + col, dbias);
 }
 
 }  // namespace
@@ -273,32 +263,39 @@ std::vector<torch::Tensor> bias_dropout_residual_ln_bwd(
   auto dw32 = acc32[0], db32 = acc32[1], dbias32 = acc32[2];
   auto stream = at::hip::getCurrentHIPStream();
   const bool drop = p > 0.0 && mask.numel() > 0;
-  const int nblocks = (int)std::min<long>((R + 3) / 4, 512);
-  TORCH_CHECK(H == 768 || H == 1024, "bdrl: hidden size must be 768 or 1024");
+  const long rows_per_chunk = 32;
+  const long chunks = (R + rows_per_chunk - 1) / rows_per_chunk;
+  const long grid = (R + 3) / 4;
   DISPATCH_FLOAT_TYPES(xsum.scalar_type(), "bdrl_bwd", [&] {
-    auto launch = [&](auto drop_c, auto nch_c) {
-      hipLaunchKernelGGL(
-          (bdrl_bwd_dx_kernel<scalar_t, decltype(drop_c)::value,
-                              decltype(nch_c)::value>),
-          dim3(nblocks), dim3(256), 0, stream,
-          (const scalar_t*)dout.data_ptr(),
-          (const scalar_t*)xsum.data_ptr(),
-          drop ? mask.data_ptr<unsigned char>() : nullptr,
-          (const scalar_t*)lnw.data_ptr(), mean.data_ptr<float>(),
-          rstd.data_ptr<float>(), (scalar_t*)dy.data_ptr(),
-          (scalar_t*)dres.data_ptr(), dw32.data_ptr<float>(),
-          db32.data_ptr<float>(), dbias32.data_ptr<float>(), H,
-          drop ? (float)p : 0.f, R, nblocks);
-    };
-    using T3 = std::integral_constant<int, 3>;
-    using T4 = std::integral_constant<int, 4>;
     if (drop) {
-      if (H == 768) launch(std::true_type{}, T3{});
-      else launch(std::true_type{}, T4{});
+      hipLaunchKernelGGL((bdrl_bwd_dx_kernel<scalar_t, true>), dim3(grid),
+                         dim3(256), 0, stream,
+                         (const scalar_t*)dout.data_ptr(),
+                         (const scalar_t*)xsum.data_ptr(),
+                         mask.data_ptr<unsigned char>(),
+                         (const scalar_t*)lnw.data_ptr(),
+                         mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                         (scalar_t*)dy.data_ptr(), (scalar_t*)dres.data_ptr(),
+                         H, (float)p, R);
     } else {
-      if (H == 768) launch(std::false_type{}, T3{});
-      else launch(std::false_type{}, T4{});
+      hipLaunchKernelGGL((bdrl_bwd_dx_kernel<scalar_t, false>), dim3(grid),
+                         dim3(256), 0, stream,
+                         (const scalar_t*)dout.data_ptr(),
+                         (const scalar_t*)xsum.data_ptr(), nullptr,
+                         (const scalar_t*)lnw.data_ptr(),
+                         mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                         (scalar_t*)dy.data_ptr(), (scalar_t*)dres.data_ptr(),
+                         H, 0.f, R);
     }
+    dim3 g2((H + 255) / 256, chunks);
+    hipLaunchKernelGGL((bdrl_bwd_dwdb_kernel<scalar_t>), g2, dim3(256), 0,
+                       stream,
+                       (const scalar_t*)dout.data_ptr(),
+                       (const scalar_t*)xsum.data_ptr(),
+                       (const scalar_t*)dy.data_ptr(),
+                       mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                       dw32.data_ptr<float>(), db32.data_ptr<float>(),
+                       dbias32.data_ptr<float>(), R, H, rows_per_chunk);
   });
   auto accT = acc32.to(lnw.scalar_type());  // one cast kernel for all three
   return {dy, accT[2], dres, accT[0], accT[1]};
