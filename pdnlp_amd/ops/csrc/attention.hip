@@ -19,12 +19,12 @@
 //    device memory; host reseeds between replays).
 //  - Backward: two atomics-free passes. dQ pass owns 64-row blocks; dK/dV
 //    pass owns 64-key blocks and computes S^T = K@Q^T so every GEMM is the
-//    native MFMA A@B^T form. Transposed B-operands (V^T, K^T, Q^T, dO^T) are
-//    gathered from the row-major LDS image with per-lane b16 reads.
+//    native MFMA A@B^T form. Transposed B-operands (V^T, K^T, Q^T, dO^T)
+//    come from ds_read_b64_tr_b16 hardware transpose reads.
 //
-// LDS image convention (shared with gemm.hip): [64 rows][128 bytes], byte
-// offset XOR-swizzled by ((row & 7) << 4); staged by global_load_lds with the
-// swizzle pre-applied to the *source* address (writes are lane-linear).
+// LDS image convention: every tile is a gfx950 "tr-image" (see tr_addr
+// below) that serves both normal and hardware-transposed fragment reads
+// from the same staging.
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
@@ -53,54 +53,99 @@ __device__ __forceinline__ unsigned short f32_bits16(float v) {
   return *reinterpret_cast<unsigned short*>(&t);
 }
 
-// stage a 64-row x 64-elem (128 B) tile into the swizzled LDS image.
-// `ld` = row stride in elements; rows clamped to max_row (reads only).
+// All attention tiles live in the gfx950 "tr-image" layout: a [64 m][64 c]
+// tile is 8 subtiles of [32 m][16 c], element (m, c) at byte
+//   2*((c&15) + (m&3)*16 + ((m>>3)&3)*64 + ((m>>2)&1)*256
+//      + ((m>>5)*4 + (c>>4))*512).
+// ONE image then serves BOTH fragment orientations:
+//  - entity-along-m fragments (row-major consumption) read 16 contiguous
+//    bytes per lane (read_n);
+//  - entity-along-c fragments (transposed consumption: V^T, K^T, Q^T, dO^T)
+//    use ds_read_b64_tr_b16 hardware transpose reads over the 128-B
+//    [4 m][16 c] blocks (read_tr4) — no per-lane b16 gathers.
+
+__device__ __forceinline__ int tr_addr(int m, int c) {
+  return 2 * ((c & 15) + (m & 3) * 16 + ((m >> 3) & 3) * 64 +
+              ((m >> 2) & 1) * 256 + ((m >> 5) * 4 + (c >> 4)) * 512);
+}
+
+// stage a [64 m][64 c] tile (row stride ld elements, rows clamped to
+// max_row) into the tr-image via global_load_lds: dest element run
+// l*8..l*8+7 decodes to (m_rel, c half); each lane fetches 16 contiguous
+// bytes of one source row — coalesced.
 template <typename T>
 __device__ __forceinline__ void stage64(const T* __restrict__ src, long ld,
                                         long row0, long max_row, char* lds) {
   const int lane = threadIdx.x & (WAVE - 1);
   const int wid = threadIdx.x >> 6;
-  const int sub_row = lane >> 3;            // 0..7
-  const int piece = lane & 7;               // 16B piece of the 128B row
-  const int kbyte = (piece * 16) ^ (sub_row << 4);
+  const int m_rel = ((lane >> 3) & 3) * 8 + (lane >> 5) * 4 + ((lane >> 1) & 3);
+  const int c0 = (lane & 1) * 8;
 #pragma unroll
-  for (int c = 0; c < 2; ++c) {
-    const int r = (wid * 2 + c) * 8 + sub_row;
-    long gr = row0 + r;
+  for (int sidx = 0; sidx < 2; ++sidx) {
+    const int sub = wid * 2 + sidx;       // subtile 0..7
+    const int s_m = sub >> 2;
+    const int s_c = sub & 3;
+    long gr = row0 + s_m * 32 + m_rel;
     gr = gr < max_row ? gr : max_row - 1;
-    const char* gp = (const char*)(src + gr * ld) + kbyte;
-    char* lp = lds + (long)(wid * 2 + c) * 8 * 128;
+    const char* gp = (const char*)(src + gr * ld + s_c * 16 + c0);
+    char* lp = lds + sub * 1024;
     __builtin_amdgcn_global_load_lds(
         (const __attribute__((address_space(1))) unsigned int*)gp,
         (__attribute__((address_space(3))) unsigned int*)lp, 16, 0, 0);
   }
 }
 
-// A/B-operand fragment: lane l holds image[row0 + (l&15)][(l>>4)*8 + i],
-// i=0..7 within the ks-th 32-element K-subtile — one ds_read_b128.
+// entity-along-m fragment: lane l holds tile[row0 + (l&15)][ks*32 +
+// (l>>4)*8 + i], i = 0..7 — one 16-B read per lane (c-run stays inside one
+// 16-c subtile).
 template <typename V8>
-__device__ __forceinline__ V8 read_frag16(const char* lds, int row0, int ks) {
+__device__ __forceinline__ V8 read_n(const char* lds, int row0, int ks) {
   const int lane = threadIdx.x & (WAVE - 1);
-  const int row = row0 + (lane & 15);
-  const int colbyte = (ks * 64 + (lane >> 4) * 16) ^ ((row & 7) << 4);
-  return *reinterpret_cast<const V8*>(lds + row * 128 + colbyte);
+  return *reinterpret_cast<const V8*>(
+      lds + tr_addr(row0 + (lane & 15), ks * 32 + (lane >> 4) * 8));
 }
 
-// Transposed fragment: B-operand rows are the image's COLUMNS. Lane l holds
-// image[key0 + (l>>4)*8 + i][d0 + (l&15)] — eight b16 gathers per fragment.
-template <typename V8>
-__device__ __forceinline__ V8 read_frag_t(const char* lds, int key0, int d0) {
+// per-lane base for a hardware-transpose fragment at (ent0 on c, m_sub):
+// lane-LINEAR 8-B addresses over each quarter's 128-B block.
+__device__ __forceinline__ unsigned int tr_base(const char* lds, int ent0,
+                                                int m_sub) {
   const int lane = threadIdx.x & (WAVE - 1);
-  const int dbyte = (d0 + (lane & 15)) * 2;
-  V8 out;
-#pragma unroll
-  for (int i = 0; i < 8; ++i) {
-    const int row = key0 + (lane >> 4) * 8 + i;
-    reinterpret_cast<unsigned short*>(&out)[i] =
-        *reinterpret_cast<const unsigned short*>(
-            lds + row * 128 + (dbyte ^ ((row & 7) << 4)));
-  }
-  return out;
+  const int sub = (m_sub >> 5) * 4 + (ent0 >> 4);
+  return (unsigned int)(unsigned long)lds + sub * 1024 + (lane & 15) * 8 +
+         (lane >> 4) * 128;
+}
+
+// four transposed fragments in ONE asm block (8x ds_read_b64_tr_b16 +
+// s_waitcnt INSIDE — the compiler cannot count asm ds ops, and the outputs
+// must be earlyclobber so they never alias the address inputs).
+typedef unsigned int uint2v __attribute__((ext_vector_type(2)));
+template <typename V8>
+__device__ __forceinline__ void read_tr4(unsigned int b0, unsigned int b1,
+                                         unsigned int b2, unsigned int b3,
+                                         V8* f) {
+  uint2v r0, r1, r2, r3, r4, r5, r6, r7;
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %8\n\t"
+      "ds_read_b64_tr_b16 %1, %8 offset:512\n\t"
+      "ds_read_b64_tr_b16 %2, %9\n\t"
+      "ds_read_b64_tr_b16 %3, %9 offset:512\n\t"
+      "ds_read_b64_tr_b16 %4, %10\n\t"
+      "ds_read_b64_tr_b16 %5, %10 offset:512\n\t"
+      "ds_read_b64_tr_b16 %6, %11\n\t"
+      "ds_read_b64_tr_b16 %7, %11 offset:512\n\t"
+      "s_waitcnt lgkmcnt(0)"
+      : "=&v"(r0), "=&v"(r1), "=&v"(r2), "=&v"(r3), "=&v"(r4), "=&v"(r5),
+        "=&v"(r6), "=&v"(r7)
+      : "v"(b0), "v"(b1), "v"(b2), "v"(b3)
+      : "memory");
+  reinterpret_cast<uint2v*>(&f[0])[0] = r0;
+  reinterpret_cast<uint2v*>(&f[0])[1] = r1;
+  reinterpret_cast<uint2v*>(&f[1])[0] = r2;
+  reinterpret_cast<uint2v*>(&f[1])[1] = r3;
+  reinterpret_cast<uint2v*>(&f[2])[0] = r4;
+  reinterpret_cast<uint2v*>(&f[2])[1] = r5;
+  reinterpret_cast<uint2v*>(&f[3])[0] = r6;
+  reinterpret_cast<uint2v*>(&f[3])[1] = r7;
 }
 
 // 16-lane (quarter-wave) butterfly reductions — score rows live across the
@@ -168,11 +213,11 @@ void fa_fwd_kernel(const T* __restrict__ qkv, const T* __restrict__ mask,
     f32x4 acc_s[4] = {};
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
-      V8 a = read_frag16<V8>(q_lds, wr, ks);
+      V8 a = read_n<V8>(q_lds, wr, ks);
 #pragma unroll
       for (int j = 0; j < 4; ++j)
         acc_s[j] =
-            mfma16<V8>(a, read_frag16<V8>(k_lds[cur], j * 16, ks), acc_s[j]);
+            mfma16<V8>(a, read_n<V8>(k_lds[cur], j * 16, ks), acc_s[j]);
     }
     float mv[4];
 #pragma unroll
@@ -216,18 +261,22 @@ void fa_fwd_kernel(const T* __restrict__ qkv, const T* __restrict__ mask,
           pv = (rr * 2.3283064365386963e-10f) >= p ? pv * inv_keep : 0.f;
         }
         *reinterpret_cast<unsigned short*>(
-            p_lds + prow * 128 + ((kcol * 2) ^ ((prow & 7) << 4))) =
+            p_lds + tr_addr(prow, kcol)) =
             f32_bits16<T>(pv);
       }
     }
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
-      V8 a = read_frag16<V8>(p_lds, wr, ks);
+      V8 a = read_n<V8>(p_lds, wr, ks);
+      V8 bv[4];
+      read_tr4<V8>(tr_base(v_lds[cur], 0, ks * 32),
+                   tr_base(v_lds[cur], 16, ks * 32),
+                   tr_base(v_lds[cur], 32, ks * 32),
+                   tr_base(v_lds[cur], 48, ks * 32), bv);
 #pragma unroll
       for (int jd = 0; jd < 4; ++jd)
-        acc_o[jd] = mfma16<V8>(a, read_frag_t<V8>(v_lds[cur], ks * 32, jd * 16),
-                               acc_o[jd]);
+        acc_o[jd] = mfma16<V8>(a, bv[jd], acc_o[jd]);
     }
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __syncthreads();
@@ -329,14 +378,14 @@ void fa_bwd_dq_kernel(const T* __restrict__ dout, const T* __restrict__ qkv,
     f32x4 acc_s[4] = {}, acc_dp[4] = {};
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
-      V8 aq = read_frag16<V8>(q_lds, wr, ks);
-      V8 ad = read_frag16<V8>(do_lds, wr, ks);
+      V8 aq = read_n<V8>(q_lds, wr, ks);
+      V8 ad = read_n<V8>(do_lds, wr, ks);
 #pragma unroll
       for (int j = 0; j < 4; ++j) {
         acc_s[j] =
-            mfma16<V8>(aq, read_frag16<V8>(k_lds[cur], j * 16, ks), acc_s[j]);
+            mfma16<V8>(aq, read_n<V8>(k_lds[cur], j * 16, ks), acc_s[j]);
         acc_dp[j] =
-            mfma16<V8>(ad, read_frag16<V8>(v_lds[cur], j * 16, ks), acc_dp[j]);
+            mfma16<V8>(ad, read_n<V8>(v_lds[cur], j * 16, ks), acc_dp[j]);
       }
     }
     float mv[4];
@@ -360,18 +409,22 @@ void fa_bwd_dq_kernel(const T* __restrict__ dout, const T* __restrict__ qkv,
         }
         const float dsv = pv * (dp - dvec_r[r]) * scale;
         *reinterpret_cast<unsigned short*>(
-            ds_lds + prow * 128 + ((kcol * 2) ^ ((prow & 7) << 4))) =
+            ds_lds + tr_addr(prow, kcol)) =
             f32_bits16<T>(dsv);
       }
     }
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
-      V8 a = read_frag16<V8>(ds_lds, wr, ks);
+      V8 a = read_n<V8>(ds_lds, wr, ks);
+      V8 bk[4];
+      read_tr4<V8>(tr_base(k_lds[cur], 0, ks * 32),
+                   tr_base(k_lds[cur], 16, ks * 32),
+                   tr_base(k_lds[cur], 32, ks * 32),
+                   tr_base(k_lds[cur], 48, ks * 32), bk);
 #pragma unroll
       for (int jd = 0; jd < 4; ++jd)
-        acc_dq[jd] = mfma16<V8>(
-            a, read_frag_t<V8>(k_lds[cur], ks * 32, jd * 16), acc_dq[jd]);
+        acc_dq[jd] = mfma16<V8>(a, bk[jd], acc_dq[jd]);
     }
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __syncthreads();
@@ -450,14 +503,14 @@ void fa_bwd_dkv_kernel(const T* __restrict__ dout, const T* __restrict__ qkv,
     f32x4 acc_st[4] = {}, acc_dpt[4] = {};
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
-      V8 ak = read_frag16<V8>(k_lds, kr, ks);
-      V8 av = read_frag16<V8>(v_lds, kr, ks);
+      V8 ak = read_n<V8>(k_lds, kr, ks);
+      V8 av = read_n<V8>(v_lds, kr, ks);
 #pragma unroll
       for (int j = 0; j < 4; ++j) {
         acc_st[j] = mfma16<V8>(
-            ak, read_frag16<V8>(q_lds[cur], j * 16, ks), acc_st[j]);
+            ak, read_n<V8>(q_lds[cur], j * 16, ks), acc_st[j]);
         acc_dpt[j] = mfma16<V8>(
-            av, read_frag16<V8>(do_lds[cur], j * 16, ks), acc_dpt[j]);
+            av, read_n<V8>(do_lds[cur], j * 16, ks), acc_dpt[j]);
       }
     }
     float pt[4][4];
@@ -487,18 +540,22 @@ void fa_bwd_dkv_kernel(const T* __restrict__ dout, const T* __restrict__ qkv,
         const int qcol = j * 16 + (lane & 15);
         const float pv = live[j][r] ? pt[j][r] * inv_keep : 0.f;
         *reinterpret_cast<unsigned short*>(
-            pds_lds + prow * 128 + ((qcol * 2) ^ ((prow & 7) << 4))) =
+            pds_lds + tr_addr(prow, qcol)) =
             f32_bits16<T>(pv);
       }
     }
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
-      V8 a = read_frag16<V8>(pds_lds, kr, ks);
+      V8 a = read_n<V8>(pds_lds, kr, ks);
+      V8 bd[4];
+      read_tr4<V8>(tr_base(do_lds[cur], 0, ks * 32),
+                   tr_base(do_lds[cur], 16, ks * 32),
+                   tr_base(do_lds[cur], 32, ks * 32),
+                   tr_base(do_lds[cur], 48, ks * 32), bd);
 #pragma unroll
       for (int jd = 0; jd < 4; ++jd)
-        acc_dv[jd] = mfma16<V8>(
-            a, read_frag_t<V8>(do_lds[cur], ks * 32, jd * 16), acc_dv[jd]);
+        acc_dv[jd] = mfma16<V8>(a, bd[jd], acc_dv[jd]);
     }
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     // dS^T tile -> LDS (same buffer); dK += dS^T @ Q
@@ -513,18 +570,22 @@ void fa_bwd_dkv_kernel(const T* __restrict__ dout, const T* __restrict__ qkv,
         const float dsv = pt[j][r] * (dpt - dvec_j) * scale;
         const int qcol = j * 16 + (lane & 15);
         *reinterpret_cast<unsigned short*>(
-            pds_lds + prow * 128 + ((qcol * 2) ^ ((prow & 7) << 4))) =
+            pds_lds + tr_addr(prow, qcol)) =
             f32_bits16<T>(dsv);
       }
     }
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
-      V8 a = read_frag16<V8>(pds_lds, kr, ks);
+      V8 a = read_n<V8>(pds_lds, kr, ks);
+      V8 bq[4];
+      read_tr4<V8>(tr_base(q_lds[cur], 0, ks * 32),
+                   tr_base(q_lds[cur], 16, ks * 32),
+                   tr_base(q_lds[cur], 32, ks * 32),
+                   tr_base(q_lds[cur], 48, ks * 32), bq);
 #pragma unroll
       for (int jd = 0; jd < 4; ++jd)
-        acc_dk[jd] = mfma16<V8>(
-            a, read_frag_t<V8>(q_lds[cur], ks * 32, jd * 16), acc_dk[jd]);
+        acc_dk[jd] = mfma16<V8>(a, bq[jd], acc_dk[jd]);
     }
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __syncthreads();

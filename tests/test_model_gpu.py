@@ -109,3 +109,32 @@ def test_grad_scaler_fp16_gpu():
     scaler.step(opt)
     scaler.update()
     assert torch.isfinite(out.loss).item()
+
+
+def test_bert_base_fp16_amp_step_gpu():
+    """fp16 AMP on the REAL kernel path (BERT-base: fp16 flash attention,
+    fp16 GEMMs/epilogues, GradScaler unscale + fused AdamW with masters)."""
+    from pdnlp_amd.amp import GradScaler
+    from pdnlp_amd.config import BertConfig
+    from pdnlp_amd.models import BertForSequenceClassification
+    from pdnlp_amd.ops.adamw import build_optimizer
+    from pdnlp_amd.utils import set_seed
+    set_seed(7)
+    cfg = BertConfig.bert_base_chinese()
+    model = BertForSequenceClassification(cfg).to(torch.float16).to(DEV)
+    model.train()
+    opt = build_optimizer(model, lr=3e-5)
+    scaler = GradScaler(init_scale=2.0 ** 12)
+    ids, mask, type_ids, labels = _batch(cfg.vocab_size, cfg.num_labels,
+                                         B=8, S=128, seed=3)
+    losses = []
+    for _ in range(4):
+        out = model(ids.to(DEV), mask.to(DEV), type_ids.to(DEV),
+                    labels.to(DEV))
+        scaler.scale(out.loss).backward()
+        scaler.step(opt)
+        scaler.update()
+        opt.zero_grad(set_to_none=True)
+        losses.append(out.loss.item())
+    assert all(torch.isfinite(torch.tensor(losses))), losses
+    assert losses[-1] < losses[0] + 0.05, losses
