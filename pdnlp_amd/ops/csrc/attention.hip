@@ -172,6 +172,7 @@ void fa_fwd_kernel(const T* __restrict__ qkv, const T* __restrict__ mask,
                    const unsigned long long* __restrict__ seed_base,
                    unsigned long long salt, float scale, float p,
                    int nh, int S) {
+  scale *= 1.4426950408889634f;  // exp2 domain (log2 e)
   const int bh = blockIdx.y;
   const long b = bh / nh, h = bh % nh;
   const long rb = blockIdx.x;
@@ -223,7 +224,8 @@ void fa_fwd_kernel(const T* __restrict__ qkv, const T* __restrict__ mask,
 #pragma unroll
     for (int j = 0; j < 4; ++j)
       mv[j] = HAS_MASK
-                  ? to_f32<T>(mask[b * S + t * 64 + j * 16 + (lane & 15)])
+                  ? to_f32<T>(mask[b * S + t * 64 + j * 16 + (lane & 15)]) *
+                        1.4426950408889634f
                   : 0.f;
     float s[4][4];
 #pragma unroll
@@ -235,12 +237,12 @@ void fa_fwd_kernel(const T* __restrict__ qkv, const T* __restrict__ mask,
         x = fmaxf(x, s[j][r]);
       }
       const float mn = fmaxf(m[r], qmax(x));
-      const float alpha = expf(m[r] - mn);   // first tile: exp(-inf) = 0
+      const float alpha = exp2f(m[r] - mn);  // first tile: exp2(-inf) = 0
       m[r] = mn;
       float sum = 0.f;
 #pragma unroll
       for (int j = 0; j < 4; ++j) {
-        s[j][r] = expf(s[j][r] - mn);
+        s[j][r] = exp2f(s[j][r] - mn);
         sum += s[j][r];
       }
       l[r] = l[r] * alpha + qsum(sum);
@@ -293,7 +295,7 @@ void fa_fwd_kernel(const T* __restrict__ qkv, const T* __restrict__ mask,
     for (int jd = 0; jd < 4; ++jd)
       obase[grow * H + jd * 16 + ccol] = from_f32<T>(acc_o[jd][r] * invl);
     if (ccol == 0)
-      lse[(long)bh * S + grow] = m[r] + logf(fmaxf(l[r], 1e-30f));
+      lse[(long)bh * S + grow] = m[r] + log2f(fmaxf(l[r], 1e-30f));
   }
 }
 
@@ -330,6 +332,7 @@ void fa_bwd_dq_kernel(const T* __restrict__ dout, const T* __restrict__ qkv,
                       const unsigned long long* __restrict__ seed_base,
                       unsigned long long salt, float scale, float p,
                       int nh, int S) {
+  const float scale2 = scale * 1.4426950408889634f;  // exp2 domain
   const int bh = blockIdx.y;
   const long b = bh / nh, h = bh % nh;
   const long rb = blockIdx.x;
@@ -392,7 +395,8 @@ void fa_bwd_dq_kernel(const T* __restrict__ dout, const T* __restrict__ qkv,
 #pragma unroll
     for (int j = 0; j < 4; ++j)
       mv[j] = HAS_MASK
-                  ? to_f32<T>(mask[b * S + t * 64 + j * 16 + (lane & 15)])
+                  ? to_f32<T>(mask[b * S + t * 64 + j * 16 + (lane & 15)]) *
+                        1.4426950408889634f
                   : 0.f;
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
@@ -401,7 +405,7 @@ void fa_bwd_dq_kernel(const T* __restrict__ dout, const T* __restrict__ qkv,
 #pragma unroll
       for (int j = 0; j < 4; ++j) {
         const int kcol = j * 16 + (lane & 15);
-        const float pv = expf(acc_s[j][r] * scale + mv[j] - lse_r[r]);
+        const float pv = exp2f(acc_s[j][r] * scale2 + mv[j] - lse_r[r]);
         float dp = acc_dp[j][r];
         if (DROP) {
           const unsigned int rr = hash_rng(seed, grow * S + t * 64 + kcol);
@@ -454,6 +458,7 @@ void fa_bwd_dkv_kernel(const T* __restrict__ dout, const T* __restrict__ qkv,
                        const unsigned long long* __restrict__ seed_base,
                        unsigned long long salt, float scale, float p,
                        int nh, int S) {
+  const float scale2 = scale * 1.4426950408889634f;  // exp2 domain
   const int bh = blockIdx.y;
   const long b = bh / nh, h = bh % nh;
   const long kb = blockIdx.x;
@@ -488,7 +493,9 @@ void fa_bwd_dkv_kernel(const T* __restrict__ dout, const T* __restrict__ qkv,
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     const long gkey = kb * 64 + kr + (lane >> 4) * 4 + r;
-    mv[r] = HAS_MASK ? to_f32<T>(mask[b * S + gkey]) : 0.f;
+    mv[r] = HAS_MASK
+                ? to_f32<T>(mask[b * S + gkey]) * 1.4426950408889634f
+                : 0.f;
   }
 
   f32x4 acc_dk[4] = {}, acc_dv[4] = {};
@@ -522,7 +529,7 @@ void fa_bwd_dkv_kernel(const T* __restrict__ dout, const T* __restrict__ qkv,
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const long gkey = kb * 64 + kr + (lane >> 4) * 4 + r;
-        pt[j][r] = expf(acc_st[j][r] * scale + mv[r] - lse_j);
+        pt[j][r] = exp2f(acc_st[j][r] * scale2 + mv[r] - lse_j);
         live[j][r] = true;
         if (DROP) {
           const unsigned int rr = hash_rng(

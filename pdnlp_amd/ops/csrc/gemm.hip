@@ -53,7 +53,7 @@ __device__ __forceinline__ float gelu_f2(float x) {
 // global_load_lds. Linear LDS image [ROWS][128 bytes]; the XOR swizzle is
 // pre-applied on the source byte offset. Each wave-instruction moves 8 rows
 // (8 lanes of 16 B per row); 4 waves x (ROWS/32) calls cover ROWS rows.
-template <typename T, int ROWS>
+template <typename T, int ROWS, int NW>
 __device__ __forceinline__ void stage_tile(const T* __restrict__ src, long ld,
                                            long row0, long max_row, long k0,
                                            char* lds) {
@@ -62,15 +62,16 @@ __device__ __forceinline__ void stage_tile(const T* __restrict__ src, long ld,
   const int sub_row = lane >> 3;           // 0..7
   const int piece = lane & 7;              // 16B piece within the 128B row
   const int kbyte = (piece * 16) ^ (sub_row << 4);  // source pre-swizzle
+  constexpr int CPW = ROWS / (8 * NW);     // 8-row chunks per wave
 #pragma unroll
-  for (int c = 0; c < ROWS / 32; ++c) {
-    const int r = (wid * (ROWS / 32) + c) * 8 + sub_row;
+  for (int c = 0; c < CPW; ++c) {
+    const int r = (wid * CPW + c) * 8 + sub_row;
     long gr = row0 + r;
     gr = gr < max_row ? gr : max_row - 1;        // clamp tail (stores guard)
     const char* gp = (const char*)(src + gr * ld + k0) + kbyte;
     // LDS dest operand is WAVE-UNIFORM (start of this call's 8-row chunk);
     // hardware writes lane l at dest + l*16 = row (l>>3), piece (l&7).
-    char* lp = lds + (long)(wid * (ROWS / 32) + c) * 8 * 128;
+    char* lp = lds + (long)(wid * CPW + c) * 8 * 128;
     __builtin_amdgcn_global_load_lds(
         (const __attribute__((address_space(1))) unsigned int*)gp,
         (__attribute__((address_space(3))) unsigned int*)lp, 16, 0, 0);
@@ -90,15 +91,16 @@ __device__ __forceinline__ V8 read_frag(const char* lds, int frag_row0,
 enum Act { ACT_NONE = 0, ACT_GELU = 1, ACT_TANH = 2 };
 
 template <typename T, typename V8, bool HAS_BIAS, int ACT, bool SAVE_PRE,
-          int BM, int BN>
-__global__ __launch_bounds__(NTHREADS)
+          int BM, int BN, int NW>
+__global__ __launch_bounds__(NW * WAVE)
 void gemm_nt_kernel(const T* __restrict__ A, const T* __restrict__ W,
                     const T* __restrict__ bias, T* __restrict__ C,
                     T* __restrict__ pre, long M, long N, long K,
                     int tiles_n, int nwg) {
-  // wave grid: 2x2 for square-ish tiles, 1x4 / 4x1 for skinny ones
-  constexpr int WM = (BM >= 128 || BN < 128) ? 2 : 1;
-  constexpr int WN = 4 / WM;
+  // wave grid: 2x2 (4 waves) for square-ish tiles, 1x4 / 4x1 for skinny
+  // ones; 2x4 at 8 waves (more waves hide the end-of-tile vmcnt stall)
+  constexpr int WM = NW == 8 ? 2 : ((BM >= 128 || BN < 128) ? 2 : 1);
+  constexpr int WN = NW / WM;
   constexpr int TM = BM / WM, TN = BN / WN;     // per-wave sub-tile
   constexpr int RM = TM / 16, RN = TN / 16;     // fragment repeats
 
@@ -122,8 +124,8 @@ void gemm_nt_kernel(const T* __restrict__ A, const T* __restrict__ W,
 
   f32x4 acc[RM][RN] = {};
 
-  stage_tile<T, BM>(A, K, m0, M, 0, lds_a[0]);
-  stage_tile<T, BN>(W, K, n0, N, 0, lds_b[0]);
+  stage_tile<T, BM, NW>(A, K, m0, M, 0, lds_a[0]);
+  stage_tile<T, BN, NW>(W, K, n0, N, 0, lds_b[0]);
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
 
@@ -131,8 +133,8 @@ void gemm_nt_kernel(const T* __restrict__ A, const T* __restrict__ W,
   int cur = 0;
   for (int t = 0; t < ntiles; ++t) {
     if (t + 1 < ntiles) {
-      stage_tile<T, BM>(A, K, m0, M, (long)(t + 1) * BK, lds_a[cur ^ 1]);
-      stage_tile<T, BN>(W, K, n0, N, (long)(t + 1) * BK, lds_b[cur ^ 1]);
+      stage_tile<T, BM, NW>(A, K, m0, M, (long)(t + 1) * BK, lds_a[cur ^ 1]);
+      stage_tile<T, BN, NW>(W, K, n0, N, (long)(t + 1) * BK, lds_b[cur ^ 1]);
     }
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
@@ -223,17 +225,21 @@ void launch_gemm(const torch::Tensor& A, const torch::Tensor& W,
   const T* bptr = has_bias ? (const T*)bias.data_ptr() : nullptr;
   T* pptr = save_pre ? (T*)pre.data_ptr() : nullptr;
 
-#define LAUNCH_T(HB, ACTV, SP, BMV, BNV)                                       \
-  hipLaunchKernelGGL((gemm_nt_kernel<T, V8, HB, ACTV, SP, BMV, BNV>),          \
-                     dim3(nwg), dim3(NTHREADS), 0, stream,                     \
+  const bool w8 = std::getenv("PDNLP_GEMM_W8") != nullptr;
+#define LAUNCH_T(HB, ACTV, SP, BMV, BNV, NWV)                                  \
+  hipLaunchKernelGGL((gemm_nt_kernel<T, V8, HB, ACTV, SP, BMV, BNV, NWV>),     \
+                     dim3(nwg), dim3(NWV * WAVE), 0, stream,                   \
                      (const T*)A.data_ptr(), (const T*)W.data_ptr(), bptr,     \
                      (T*)C.data_ptr(), pptr, M, N, K, tiles_n, nwg)
 #define LAUNCH(HB, ACTV, SP)                                                   \
   do {                                                                         \
-    if (tc.bm == 64 && tc.bn == 128) LAUNCH_T(HB, ACTV, SP, 64, 128);          \
-    else if (tc.bm == 128 && tc.bn == 64) LAUNCH_T(HB, ACTV, SP, 128, 64);     \
-    else if (tc.bm == 64 && tc.bn == 64) LAUNCH_T(HB, ACTV, SP, 64, 64);       \
-    else LAUNCH_T(HB, ACTV, SP, 128, 128);                                     \
+    if (tc.bm == 64 && tc.bn == 128) {                                         \
+      if (w8) LAUNCH_T(HB, ACTV, SP, 64, 128, 8);                              \
+      else LAUNCH_T(HB, ACTV, SP, 64, 128, 4);                                 \
+    } else if (tc.bm == 128 && tc.bn == 64) LAUNCH_T(HB, ACTV, SP, 128, 64, 4);\
+    else if (tc.bm == 64 && tc.bn == 64) LAUNCH_T(HB, ACTV, SP, 64, 64, 4);    \
+    else if (w8) LAUNCH_T(HB, ACTV, SP, 128, 128, 8);                          \
+    else LAUNCH_T(HB, ACTV, SP, 128, 128, 4);                                  \
   } while (0)
 
   if (act == ACT_NONE) {
