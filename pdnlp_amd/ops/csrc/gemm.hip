@@ -225,7 +225,8 @@ void launch_gemm(const torch::Tensor& A, const torch::Tensor& W,
   const T* bptr = has_bias ? (const T*)bias.data_ptr() : nullptr;
   T* pptr = save_pre ? (T*)pre.data_ptr() : nullptr;
 
-  const bool w8 = std::getenv("PDNLP_GEMM_W8") != nullptr;
+  // 8 waves default (swept +8-15% over 4): PDNLP_GEMM_W4 reverts
+  const bool w8 = std::getenv("PDNLP_GEMM_W4") == nullptr;
 #define LAUNCH_T(HB, ACTV, SP, BMV, BNV, NWV)                                  \
   hipLaunchKernelGGL((gemm_nt_kernel<T, V8, HB, ACTV, SP, BMV, BNV, NWV>),     \
                      dim3(nwg), dim3(NWV * WAVE), 0, stream,                   \
