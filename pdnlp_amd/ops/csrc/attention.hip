@@ -191,6 +191,7 @@ void fa_fwd_kernel(const T* __restrict__ qkv, const T* __restrict__ mask,
   const int wid = threadIdx.x >> 6;
   const int wr = wid * 16;
   const unsigned long long seed = DROP ? (*seed_base + salt) : 0ull;
+  const unsigned int seed32 = (unsigned int)(seed ^ (seed >> 32));
   const float inv_keep = DROP ? 1.f / (1.f - p) : 1.f;
 
   stage64<T>(qbase, ld, rb * 64, S, q_lds);
@@ -253,13 +254,13 @@ void fa_fwd_kernel(const T* __restrict__ qkv, const T* __restrict__ mask,
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int prow = wr + (lane >> 4) * 4 + r;
-      const unsigned long long grow = (unsigned long long)bh * S + rb * 64 + prow;
+      const unsigned int grow = (unsigned int)bh * S + rb * 64 + prow;
 #pragma unroll
       for (int j = 0; j < 4; ++j) {
         const int kcol = j * 16 + (lane & 15);
         float pv = s[j][r];
         if (DROP) {
-          const unsigned int rr = hash_rng(seed, grow * S + t * 64 + kcol);
+          const unsigned int rr = hash_rng32(seed32, grow * S + t * 64 + kcol);
           pv = (rr * 2.3283064365386963e-10f) >= p ? pv * inv_keep : 0.f;
         }
         *reinterpret_cast<unsigned short*>(
@@ -353,6 +354,7 @@ void fa_bwd_dq_kernel(const T* __restrict__ dout, const T* __restrict__ qkv,
   const int wid = threadIdx.x >> 6;
   const int wr = wid * 16;
   const unsigned long long seed = DROP ? (*seed_base + salt) : 0ull;
+  const unsigned int seed32 = (unsigned int)(seed ^ (seed >> 32));
   const float inv_keep = DROP ? 1.f / (1.f - p) : 1.f;
 
   stage64<T>(qbase, ld, rb * 64, S, q_lds);
@@ -401,14 +403,14 @@ void fa_bwd_dq_kernel(const T* __restrict__ dout, const T* __restrict__ qkv,
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const int prow = wr + (lane >> 4) * 4 + r;
-      const unsigned long long grow = (unsigned long long)bh * S + rb * 64 + prow;
+      const unsigned int grow = (unsigned int)bh * S + rb * 64 + prow;
 #pragma unroll
       for (int j = 0; j < 4; ++j) {
         const int kcol = j * 16 + (lane & 15);
         const float pv = exp2f(acc_s[j][r] * scale2 + mv[j] - lse_r[r]);
         float dp = acc_dp[j][r];
         if (DROP) {
-          const unsigned int rr = hash_rng(seed, grow * S + t * 64 + kcol);
+          const unsigned int rr = hash_rng32(seed32, grow * S + t * 64 + kcol);
           dp = (rr * 2.3283064365386963e-10f) >= p ? dp * inv_keep : 0.f;
         }
         const float dsv = pv * (dp - dvec_r[r]) * scale;
@@ -480,6 +482,7 @@ void fa_bwd_dkv_kernel(const T* __restrict__ dout, const T* __restrict__ qkv,
   const int wid = threadIdx.x >> 6;
   const int kr = wid * 16;
   const unsigned long long seed = DROP ? (*seed_base + salt) : 0ull;
+  const unsigned int seed32 = (unsigned int)(seed ^ (seed >> 32));
   const float inv_keep = DROP ? 1.f / (1.f - p) : 1.f;
 
   stage64<T>(kbase, ld, kb * 64, S, k_lds);
@@ -532,8 +535,9 @@ void fa_bwd_dkv_kernel(const T* __restrict__ dout, const T* __restrict__ qkv,
         pt[j][r] = exp2f(acc_st[j][r] * scale2 + mv[r] - lse_j);
         live[j][r] = true;
         if (DROP) {
-          const unsigned int rr = hash_rng(
-              seed, ((unsigned long long)bh * S + qrow) * S + gkey);
+          const unsigned int rr = hash_rng32(
+              seed32, ((unsigned int)bh * S + (unsigned int)qrow) * S +
+                          (unsigned int)gkey);
           live[j][r] = (rr * 2.3283064365386963e-10f) >= p;
         }
       }

@@ -96,3 +96,18 @@ __device__ __forceinline__ unsigned int hash_rng(unsigned long long seed,
   z = (z ^ (z >> 27)) * 0x94D049BB133111EBull;
   return (unsigned int)(z >> 32);
 }
+
+// 32-bit variant (murmur3 finalizer) for hot in-kernel mask generation:
+// 64-bit multiplies lower to mul_lo/mul_hi chains on CDNA and the fused
+// attention evaluates 16 masks per lane per KV tile. Element index must fit
+// u32 (B*nh*S*S < 2^32 for every supported config).
+__device__ __forceinline__ unsigned int hash_rng32(unsigned int seed32,
+                                                   unsigned int idx) {
+  unsigned int x = idx ^ seed32;
+  x ^= x >> 16;
+  x *= 0x7feb352du;
+  x ^= x >> 15;
+  x *= 0x846ca68bu;
+  x ^= x >> 16;
+  return x;
+}
