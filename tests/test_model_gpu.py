@@ -138,3 +138,25 @@ def test_bert_base_fp16_amp_step_gpu():
         losses.append(out.loss.item())
     assert all(torch.isfinite(torch.tensor(losses))), losses
     assert losses[-1] < losses[0] + 0.05, losses
+
+
+def test_roberta_base_gpu_step():
+    """RoBERTa-base (alt HF encoder, BASELINE config 5) one fused-path
+    fwd+bwd step on GPU."""
+    from pdnlp_amd.models import build_model
+    from pdnlp_amd.ops.adamw import build_optimizer
+    from pdnlp_amd.utils import set_seed
+    set_seed(11)
+    model = build_model("roberta-base").to(torch.bfloat16).to(DEV)
+    model.train()
+    opt = build_optimizer(model, lr=3e-5)
+    vocab = model.config.vocab_size
+    g = torch.Generator().manual_seed(0)
+    ids = torch.randint(10, vocab, (8, 128), generator=g).to(DEV)
+    mask = torch.ones_like(ids)
+    labels = torch.randint(0, 6, (8,), generator=g).to(DEV)
+    out = model(input_ids=ids, attention_mask=mask, labels=labels)
+    out.loss.backward()
+    opt.step()
+    torch.cuda.synchronize()
+    assert torch.isfinite(out.loss).item()
