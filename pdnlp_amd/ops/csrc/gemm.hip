@@ -188,6 +188,121 @@ void gemm_nt_kernel(const T* __restrict__ A, const T* __restrict__ W,
   }
 }
 
+
+// 3-deep pipelined variant (8 waves, dynamic LDS): stage tile t+2 while
+// computing tile t, wait only vmcnt(NGLDS) (= t+2's in-flight loads) before
+// a RAW s_barrier — __syncthreads() would drain the glds queue to zero
+// (the ~70% wave-park the PMC profile shows on the 2-barrier loop).
+template <typename T, typename V8, bool HAS_BIAS, int ACT, bool SAVE_PRE,
+          int BM, int BN>
+__global__ __launch_bounds__(512)
+void gemm_nt_pipe_kernel(const T* __restrict__ A, const T* __restrict__ W,
+                         const T* __restrict__ bias, T* __restrict__ C,
+                         T* __restrict__ pre, long M, long N, long K,
+                         int tiles_n, int nwg) {
+  constexpr int NW = 8;
+  constexpr int WM = 2, WN = 4;
+  constexpr int TM = BM / WM, TN = BN / WN;
+  constexpr int RM = TM / 16, RN = TN / 16;
+  constexpr int NGLDS = BM / (8 * NW) + BN / (8 * NW);  // glds per wave/tile
+
+  int wg = blockIdx.x;
+  {
+    const int nxcd = 8;
+    const int q = nwg / nxcd, r = nwg % nxcd;
+    const int xcd = wg % nxcd, idx = wg / nxcd;
+    wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  }
+  const long tile_m = wg / tiles_n, tile_n = wg % tiles_n;
+  const long m0 = tile_m * BM, n0 = tile_n * BN;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  auto lds_a = [&](int i) -> char* { return smem + i * (BM + BN) * 128; };
+  auto lds_b = [&](int i) -> char* {
+    return smem + i * (BM + BN) * 128 + BM * 128;
+  };
+
+  const int wid = threadIdx.x >> 6;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wr = (wid / WN) * TM, wc = (wid % WN) * TN;
+
+  f32x4 acc[RM][RN] = {};
+
+  const int ntiles = (int)(K / BK);
+  stage_tile<T, BM, NW>(A, K, m0, M, 0, lds_a(0));
+  stage_tile<T, BN, NW>(W, K, n0, N, 0, lds_b(0));
+  if (1 < ntiles) {
+    stage_tile<T, BM, NW>(A, K, m0, M, BK, lds_a(1));
+    stage_tile<T, BN, NW>(W, K, n0, N, BK, lds_b(1));
+  }
+  if constexpr (NGLDS == 3) asm volatile("s_waitcnt vmcnt(3)" ::: "memory");
+  else asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  for (int t = 0; t < ntiles; ++t) {
+    const int cur = t % 3;
+    const bool staged = t + 2 < ntiles;
+    if (staged) {
+      stage_tile<T, BM, NW>(A, K, m0, M, (long)(t + 2) * BK,
+                            lds_a((t + 2) % 3));
+      stage_tile<T, BN, NW>(W, K, n0, N, (long)(t + 2) * BK,
+                            lds_b((t + 2) % 3));
+    }
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      V8 a_frag[RM], b_frag[RN];
+#pragma unroll
+      for (int i = 0; i < RM; ++i)
+        a_frag[i] = read_frag<V8>(lds_a(cur), wr + i * 16, ks);
+#pragma unroll
+      for (int j = 0; j < RN; ++j)
+        b_frag[j] = read_frag<V8>(lds_b(cur), wc + j * 16, ks);
+#pragma unroll
+      for (int i = 0; i < RM; ++i) {
+#pragma unroll
+        for (int j = 0; j < RN; ++j) {
+          if constexpr (std::is_same<V8, bf16x8>::value) {
+            acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
+          } else {
+            acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_f16(
+                a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
+          }
+        }
+      }
+    }
+    if (staged) {
+      if constexpr (NGLDS == 3) asm volatile("s_waitcnt vmcnt(3)" ::: "memory");
+      else asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
+  }
+
+  const int crow_off = (lane >> 4) * 4;
+  const int ccol = lane & 15;
+#pragma unroll
+  for (int i = 0; i < RM; ++i) {
+#pragma unroll
+    for (int j = 0; j < RN; ++j) {
+      const long n = n0 + wc + j * 16 + ccol;
+      if (n >= N) continue;
+      const float bv = HAS_BIAS ? to_f32<T>(bias[n]) : 0.f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const long m = m0 + wr + i * 16 + crow_off + r;
+        if (m >= M) continue;
+        float v = acc[i][j][r] + bv;
+        if (SAVE_PRE) pre[m * N + n] = from_f32<T>(v);
+        if (ACT == ACT_GELU) v = gelu_f2(v);
+        if (ACT == ACT_TANH) v = tanhf(v);
+        C[m * N + n] = from_f32<T>(v);
+      }
+    }
+  }
+}
+
 struct TileChoice { int bm, bn; };
 
 // pick the tile so the grid fills 256 CUs (>= ~2 WGs per CU preferred),
@@ -227,6 +342,22 @@ void launch_gemm(const torch::Tensor& A, const torch::Tensor& W,
 
   // 8 waves default (swept +8-15% over 4): PDNLP_GEMM_W4 reverts
   const bool w8 = std::getenv("PDNLP_GEMM_W4") == nullptr;
+  const bool pipe = std::getenv("PDNLP_GEMM_PIPE") != nullptr;
+#define LAUNCH_P(HB, ACTV, SP, BMV, BNV)                                       \
+  do {                                                                         \
+    constexpr int shmem = 3 * (BMV + BNV) * 128;                               \
+    static bool attr_set_##BMV##_##BNV = false;                                \
+    if (!attr_set_##BMV##_##BNV) {                                             \
+      hipFuncSetAttribute(                                                     \
+          (const void*)&gemm_nt_pipe_kernel<T, V8, HB, ACTV, SP, BMV, BNV>,    \
+          hipFuncAttributeMaxDynamicSharedMemorySize, shmem);                  \
+      attr_set_##BMV##_##BNV = true;                                           \
+    }                                                                          \
+    hipLaunchKernelGGL((gemm_nt_pipe_kernel<T, V8, HB, ACTV, SP, BMV, BNV>),   \
+                       dim3(nwg), dim3(512), shmem, stream,                    \
+                       (const T*)A.data_ptr(), (const T*)W.data_ptr(), bptr,   \
+                       (T*)C.data_ptr(), pptr, M, N, K, tiles_n, nwg);         \
+  } while (0)
 #define LAUNCH_T(HB, ACTV, SP, BMV, BNV, NWV)                                  \
   hipLaunchKernelGGL((gemm_nt_kernel<T, V8, HB, ACTV, SP, BMV, BNV, NWV>),     \
                      dim3(nwg), dim3(NWV * WAVE), 0, stream,                   \
@@ -235,10 +366,12 @@ void launch_gemm(const torch::Tensor& A, const torch::Tensor& W,
 #define LAUNCH(HB, ACTV, SP)                                                   \
   do {                                                                         \
     if (tc.bm == 64 && tc.bn == 128) {                                         \
-      if (w8) LAUNCH_T(HB, ACTV, SP, 64, 128, 8);                              \
+      if (pipe) LAUNCH_P(HB, ACTV, SP, 64, 128);                               \
+      else if (w8) LAUNCH_T(HB, ACTV, SP, 64, 128, 8);                         \
       else LAUNCH_T(HB, ACTV, SP, 64, 128, 4);                                 \
     } else if (tc.bm == 128 && tc.bn == 64) LAUNCH_T(HB, ACTV, SP, 128, 64, 4);\
     else if (tc.bm == 64 && tc.bn == 64) LAUNCH_T(HB, ACTV, SP, 64, 64, 4);    \
+    else if (pipe) LAUNCH_P(HB, ACTV, SP, 128, 128);                           \
     else if (w8) LAUNCH_T(HB, ACTV, SP, 128, 128, 8);                          \
     else LAUNCH_T(HB, ACTV, SP, 128, 128, 4);                                  \
   } while (0)
