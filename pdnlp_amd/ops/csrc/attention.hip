@@ -165,7 +165,7 @@ __device__ __forceinline__ float qsum(float x) {
 // forward
 // ---------------------------------------------------------------------------
 
-template <typename T, typename V8, bool HAS_MASK, bool DROP>
+template <typename T, typename V8, bool HAS_MASK, bool DROP, bool PRELOAD>
 __global__ __launch_bounds__(NTHREADS)
 void fa_fwd_kernel(const T* __restrict__ qkv, const T* __restrict__ mask,
                    T* __restrict__ o, float* __restrict__ lse,
@@ -197,6 +197,12 @@ void fa_fwd_kernel(const T* __restrict__ qkv, const T* __restrict__ mask,
   stage64<T>(qbase, ld, rb * 64, S, q_lds);
   stage64<T>(kbase, ld, 0, S, k_lds[0]);
   stage64<T>(vbase, ld, 0, S, v_lds[0]);
+  if (PRELOAD && S > 64) {
+    // S <= 128: the whole K/V fits the double buffers — stage everything
+    // up front and run the KV loop with NO mid-loop barriers/waits
+    stage64<T>(kbase, ld, 64, S, k_lds[1]);
+    stage64<T>(vbase, ld, 64, S, v_lds[1]);
+  }
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
 
@@ -208,7 +214,7 @@ void fa_fwd_kernel(const T* __restrict__ qkv, const T* __restrict__ mask,
   const int nt = S / 64;
   int cur = 0;
   for (int t = 0; t < nt; ++t) {
-    if (t + 1 < nt) {
+    if (!PRELOAD && t + 1 < nt) {
       stage64<T>(kbase, ld, (long)(t + 1) * 64, S, k_lds[cur ^ 1]);
       stage64<T>(vbase, ld, (long)(t + 1) * 64, S, v_lds[cur ^ 1]);
     }
@@ -281,8 +287,10 @@ void fa_fwd_kernel(const T* __restrict__ qkv, const T* __restrict__ mask,
       for (int jd = 0; jd < 4; ++jd)
         acc_o[jd] = mfma16<V8>(a, bv[jd], acc_o[jd]);
     }
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    __syncthreads();
+    if (!PRELOAD) {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __syncthreads();
+    }
     cur ^= 1;
   }
 
@@ -324,7 +332,7 @@ void fa_bwd_pre_kernel(const T* __restrict__ dout, const T* __restrict__ o,
 // backward dQ: grid over 64-row blocks; recompute P from lse, stream K/V
 // ---------------------------------------------------------------------------
 
-template <typename T, typename V8, bool HAS_MASK, bool DROP>
+template <typename T, typename V8, bool HAS_MASK, bool DROP, bool PRELOAD>
 __global__ __launch_bounds__(NTHREADS)
 void fa_bwd_dq_kernel(const T* __restrict__ dout, const T* __restrict__ qkv,
                       const float* __restrict__ lse,
@@ -361,6 +369,10 @@ void fa_bwd_dq_kernel(const T* __restrict__ dout, const T* __restrict__ qkv,
   stage64<T>(dobase, H, rb * 64, S, do_lds);
   stage64<T>(kbase, ld, 0, S, k_lds[0]);
   stage64<T>(vbase, ld, 0, S, v_lds[0]);
+  if (PRELOAD && S > 64) {
+    stage64<T>(kbase, ld, 64, S, k_lds[1]);
+    stage64<T>(vbase, ld, 64, S, v_lds[1]);
+  }
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
 
@@ -376,7 +388,7 @@ void fa_bwd_dq_kernel(const T* __restrict__ dout, const T* __restrict__ qkv,
   const int nt = S / 64;
   int cur = 0;
   for (int t = 0; t < nt; ++t) {
-    if (t + 1 < nt) {
+    if (!PRELOAD && t + 1 < nt) {
       stage64<T>(kbase, ld, (long)(t + 1) * 64, S, k_lds[cur ^ 1]);
       stage64<T>(vbase, ld, (long)(t + 1) * 64, S, v_lds[cur ^ 1]);
     }
@@ -432,8 +444,10 @@ void fa_bwd_dq_kernel(const T* __restrict__ dout, const T* __restrict__ qkv,
       for (int jd = 0; jd < 4; ++jd)
         acc_dq[jd] = mfma16<V8>(a, bk[jd], acc_dq[jd]);
     }
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    __syncthreads();
+    if (!PRELOAD) {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __syncthreads();
+    }
     cur ^= 1;
   }
 
@@ -451,7 +465,7 @@ void fa_bwd_dq_kernel(const T* __restrict__ dout, const T* __restrict__ qkv,
 // backward dK/dV: grid over 64-key blocks; S^T = K@Q^T so all GEMMs are A@B^T
 // ---------------------------------------------------------------------------
 
-template <typename T, typename V8, bool HAS_MASK, bool DROP>
+template <typename T, typename V8, bool HAS_MASK, bool DROP, bool PRELOAD>
 __global__ __launch_bounds__(NTHREADS)
 void fa_bwd_dkv_kernel(const T* __restrict__ dout, const T* __restrict__ qkv,
                        const float* __restrict__ lse,
@@ -489,6 +503,10 @@ void fa_bwd_dkv_kernel(const T* __restrict__ dout, const T* __restrict__ qkv,
   stage64<T>(vbase, ld, kb * 64, S, v_lds);
   stage64<T>(qbase, ld, 0, S, q_lds[0]);
   stage64<T>(dobase, H, 0, S, do_lds[0]);
+  if (PRELOAD && S > 64) {
+    stage64<T>(qbase, ld, 64, S, q_lds[1]);
+    stage64<T>(dobase, H, 64, S, do_lds[1]);
+  }
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
 
@@ -505,7 +523,7 @@ void fa_bwd_dkv_kernel(const T* __restrict__ dout, const T* __restrict__ qkv,
   const int nt = S / 64;
   int cur = 0;
   for (int t = 0; t < nt; ++t) {
-    if (t + 1 < nt) {
+    if (!PRELOAD && t + 1 < nt) {
       stage64<T>(qbase, ld, (long)(t + 1) * 64, S, q_lds[cur ^ 1]);
       stage64<T>(dobase, H, (long)(t + 1) * 64, S, do_lds[cur ^ 1]);
     }
@@ -598,8 +616,10 @@ void fa_bwd_dkv_kernel(const T* __restrict__ dout, const T* __restrict__ qkv,
       for (int jd = 0; jd < 4; ++jd)
         acc_dk[jd] = mfma16<V8>(a, bq[jd], acc_dk[jd]);
     }
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    __syncthreads();
+    if (!PRELOAD) {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __syncthreads();
+    }
     cur ^= 1;
   }
 
@@ -642,7 +662,16 @@ static void check_fa_args(const torch::Tensor& qkv, const torch::Tensor& mask,
 // launch macros live at file scope (a #define cannot appear inside a macro
 // argument — DISPATCH_FLOAT_TYPES takes the body as one)
 #define FA_FWD(HM, DR)                                                         \
-  hipLaunchKernelGGL((fa_fwd_kernel<scalar_t, V8, HM, DR>), grid,              \
+  if (S <= 128)                                                                \
+    hipLaunchKernelGGL((fa_fwd_kernel<scalar_t, V8, HM, DR, true>), grid,      \
+                       dim3(NTHREADS), 0, stream,                              \
+                       (const scalar_t*)qkv.data_ptr(),                        \
+                       has_mask ? (const scalar_t*)mask.data_ptr() : nullptr,  \
+                       (scalar_t*)o.data_ptr(), (float*)lse.data_ptr(),        \
+                       seed_ptr, (unsigned long long)salt, (float)scale,       \
+                       (float)p, (int)nh, (int)S);                             \
+  else                                                                         \
+    hipLaunchKernelGGL((fa_fwd_kernel<scalar_t, V8, HM, DR, false>), grid,     \
                      dim3(NTHREADS), 0, stream,                                \
                      (const scalar_t*)qkv.data_ptr(),                          \
                      has_mask ? (const scalar_t*)mask.data_ptr() : nullptr,    \
@@ -651,7 +680,20 @@ static void check_fa_args(const torch::Tensor& qkv, const torch::Tensor& mask,
                      (float)p, (int)nh, (int)S)
 
 #define FA_BWD(KERN, HM, DR)                                                   \
-  hipLaunchKernelGGL((KERN<scalar_t, V8, HM, DR>), grid, dim3(NTHREADS), 0,    \
+  if (S <= 128)                                                                \
+    hipLaunchKernelGGL((KERN<scalar_t, V8, HM, DR, true>), grid,               \
+                       dim3(NTHREADS), 0, stream,                              \
+                       (const scalar_t*)dout.data_ptr(),                       \
+                       (const scalar_t*)qkv.data_ptr(),                        \
+                       (const float*)lse.data_ptr(),                           \
+                       (const float*)dvec.data_ptr(),                          \
+                       has_mask ? (const scalar_t*)mask.data_ptr() : nullptr,  \
+                       (scalar_t*)dqkv.data_ptr(), seed_ptr,                   \
+                       (unsigned long long)salt, (float)scale, (float)p,       \
+                       (int)nh, (int)S);                                       \
+  else                                                                         \
+    hipLaunchKernelGGL((KERN<scalar_t, V8, HM, DR, false>), grid,              \
+                       dim3(NTHREADS), 0,                                      \
                      stream, (const scalar_t*)dout.data_ptr(),                 \
                      (const scalar_t*)qkv.data_ptr(),                          \
                      (const float*)lse.data_ptr(),                             \
