@@ -152,11 +152,12 @@ class Trainer:
                 stepped = self._backward_and_step(loss, accum_boundary)
                 self.global_step += 1
                 timer.step(labels.shape[0] * self.world)
-                if self.global_step % args.loss_reduce_every == 0:
-                    printed_loss = self.loss_reduce(loss).item()
-                else:
-                    printed_loss = loss.item()
                 if self.global_step % args.log_every == 0:
+                    # the .item() device sync happens only on logged steps
+                    if self.global_step % args.loss_reduce_every == 0:
+                        printed_loss = self.loss_reduce(loss).item()
+                    else:
+                        printed_loss = loss.item()
                     rank0_print(
                         f"【train】 epoch：{epoch}/{args.epochs} "
                         f"step：{self.global_step}/{total_step} "
