@@ -25,21 +25,29 @@ def multi_tensor_adamw(params: List[torch.Tensor], grads: List[torch.Tensor],
                        masters: List[Optional[torch.Tensor]],
                        lr: float, beta1: float, beta2: float, eps: float,
                        weight_decay: float, step: int,
-                       grad_scale_inv: float = 1.0) -> None:
-    """Apply one AdamW step to a flat list of tensors (same group)."""
+                       grad_scale_inv: float = 1.0,
+                       found_inf: Optional[torch.Tensor] = None) -> None:
+    """Apply one AdamW step to a flat list of tensors (same group).
+
+    ``found_inf``: optional fp32[1] device flag — the HIP kernel skips the
+    whole update when it is nonzero (AMP overflow, no host sync)."""
     if not params:
         return
     bc1 = 1.0 - beta1 ** step
     bc2 = 1.0 - beta2 ** step
     if hip_enabled(params[0]) and getattr(ext(), "multi_tensor_adamw", None) is not None:
         use_master = masters[0] is not None
+        finf = found_inf if found_inf is not None else torch.Tensor()
         for i in range(0, len(params), CHUNK):
             ext().multi_tensor_adamw(
                 params[i:i + CHUNK], grads[i:i + CHUNK],
                 exp_avgs[i:i + CHUNK], exp_avg_sqs[i:i + CHUNK],
                 masters[i:i + CHUNK] if use_master else [],
-                lr, beta1, beta2, eps, weight_decay, bc1, bc2, grad_scale_inv)
+                lr, beta1, beta2, eps, weight_decay, bc1, bc2, grad_scale_inv,
+                finf)
         return
+    if found_inf is not None and float(found_inf.item()) != 0.0:
+        return  # torch fallback: synchronous skip
     # torch fallback (also the numerics reference)
     with torch.no_grad():
         for p, g, m, v, mw in zip(params, grads, exp_avgs, exp_avg_sqs, masters):
@@ -63,7 +71,8 @@ class FusedAdamW(torch.optim.Optimizer):
         self.master_weights = master_weights
 
     @torch.no_grad()
-    def step(self, closure=None, grad_scale_inv: float = 1.0):
+    def step(self, closure=None, grad_scale_inv: float = 1.0,
+             found_inf: Optional[torch.Tensor] = None):
         loss = None
         if closure is not None:
             with torch.enable_grad():
@@ -94,7 +103,8 @@ class FusedAdamW(torch.optim.Optimizer):
                 b1, b2 = group["betas"]
                 multi_tensor_adamw(params, grads, ms, vs, masters,
                                    group["lr"], b1, b2, group["eps"],
-                                   group["weight_decay"], step, grad_scale_inv)
+                                   group["weight_decay"], step, grad_scale_inv,
+                                   found_inf=found_inf)
         return loss
 
 

@@ -41,7 +41,11 @@ struct AdamWMeta {
 template <typename T, bool MASTER>
 __global__ void adamw_kernel(AdamWMeta meta, float lr, float beta1,
                              float beta2, float eps, float wd, float bc1,
-                             float bc2, float grad_scale_inv) {
+                             float bc2, float grad_scale_inv,
+                             const float* __restrict__ found_inf) {
+  // AMP: skip the whole update on overflow, device-side (no host sync —
+  // the GradScaler's per-step found_inf.item() stalls the launch pipeline)
+  if (found_inf != nullptr && *found_inf != 0.f) return;
   // find tensor for this block
   int t = 0;
   while (t + 1 < meta.ntensors && blockIdx.x >= meta.block_prefix[t + 1]) ++t;
@@ -154,8 +158,11 @@ void multi_tensor_adamw(std::vector<torch::Tensor> params,
                         std::vector<torch::Tensor> masters, double lr,
                         double beta1, double beta2, double eps,
                         double weight_decay, double bc1, double bc2,
-                        double grad_scale_inv) {
+                        double grad_scale_inv, torch::Tensor found_inf) {
   TORCH_CHECK(!params.empty());
+  const float* finf = (found_inf.defined() && found_inf.numel() > 0)
+                          ? found_inf.data_ptr<float>()
+                          : nullptr;
   const bool master = !masters.empty();
   auto stream = at::hip::getCurrentHIPStream();
   const auto dtype = params[0].scalar_type();
@@ -186,13 +193,13 @@ void multi_tensor_adamw(std::vector<torch::Tensor> params,
                            dim3(BLOCK), 0, stream, meta, (float)lr,
                            (float)beta1, (float)beta2, (float)eps,
                            (float)weight_decay, (float)bc1, (float)bc2,
-                           (float)grad_scale_inv);
+                           (float)grad_scale_inv, finf);
       } else {
         hipLaunchKernelGGL((adamw_kernel<scalar_t, false>), dim3(blocks),
                            dim3(BLOCK), 0, stream, meta, (float)lr,
                            (float)beta1, (float)beta2, (float)eps,
                            (float)weight_decay, (float)bc1, (float)bc2,
-                           (float)grad_scale_inv);
+                           (float)grad_scale_inv, finf);
       }
     });
   }

@@ -48,7 +48,7 @@ void multi_tensor_adamw(std::vector<torch::Tensor> params,
                         std::vector<torch::Tensor> masters, double lr,
                         double beta1, double beta2, double eps,
                         double weight_decay, double bc1, double bc2,
-                        double grad_scale_inv);
+                        double grad_scale_inv, torch::Tensor found_inf);
 void multi_tensor_unscale(std::vector<torch::Tensor> grads,
                           torch::Tensor found_inf, double inv_scale);
 std::vector<torch::Tensor> gemm_nt_fwd(torch::Tensor A, torch::Tensor W,
