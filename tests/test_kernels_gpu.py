@@ -265,7 +265,8 @@ def test_multi_tensor_adamw(dtype):
     e = ext()
     e.multi_tensor_adamw(params, grads, ms, vs,
                          masters if use_master else [],
-                         1e-3, 0.9, 0.999, 1e-8, 0.01, 0.1, 0.001, 1.0)
+                         1e-3, 0.9, 0.999, 1e-8, 0.01, 0.1, 0.001, 1.0,
+                         torch.Tensor())
     # python reference (the fallback in ops.adamw)
     import os
     os.environ["PDNLP_FORCE_TORCH"] = "1"
@@ -454,3 +455,23 @@ def test_gemm_tn(M, N, K):
     C = e.gemm_tn(A, B)
     ref = A.float().t() @ B.float()
     torch.testing.assert_close(C.float(), ref, rtol=3e-2, atol=3e-1)
+
+
+def test_adamw_device_side_overflow_skip():
+    """found_inf flag set on device -> the fused AdamW must be a no-op."""
+    e = ext()
+    torch.manual_seed(9)
+    p = torch.randn(1000, device=DEV, dtype=torch.bfloat16)
+    p0 = p.clone()
+    g = torch.randn_like(p)
+    m = torch.zeros(1000, device=DEV, dtype=torch.float32)
+    v = torch.zeros_like(m)
+    flag = torch.ones(1, device=DEV, dtype=torch.float32)
+    e.multi_tensor_adamw([p], [g], [m], [v], [], 1e-2, 0.9, 0.999, 1e-8,
+                         0.0, 0.1, 0.001, 1.0, flag)
+    torch.testing.assert_close(p, p0)   # skipped
+    assert m.abs().max().item() == 0.0
+    flag.zero_()
+    e.multi_tensor_adamw([p], [g], [m], [v], [], 1e-2, 0.9, 0.999, 1e-8,
+                         0.0, 0.1, 0.001, 1.0, flag)
+    assert not torch.equal(p, p0)       # applied
