@@ -146,3 +146,9 @@ def _ddp_training_parity(rank, world):
 
 def test_ddp_training_parity():
     run_distributed(_ddp_training_parity, world=2)
+
+
+def test_ddp_grads_match_reference_world4():
+    """8-GPU-shaped sanity at world=4 (gloo): bucket launch order and
+    views stay correct as the bucket count interacts with more ranks."""
+    run_distributed(_ddp_grads_match_reference, world=4)

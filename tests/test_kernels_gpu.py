@@ -334,17 +334,18 @@ def _attn_ref(qkv, mask, nh, p=0.0):
 
 @pytest.mark.parametrize("S,nh", [(128, 12), (512, 4), (64, 2)])
 @pytest.mark.parametrize("with_mask", [True, False])
-def test_flash_attn_fwd(S, nh, with_mask):
+@pytest.mark.parametrize("dtype", [torch.bfloat16, torch.float16])
+def test_flash_attn_fwd(S, nh, with_mask, dtype):
     e = ext()
     torch.manual_seed(0)
     B, hd = 3, 64
     H = nh * hd
-    qkv = torch.randn(B, S, 3 * H, device=DEV, dtype=torch.bfloat16)
+    qkv = torch.randn(B, S, 3 * H, device=DEV, dtype=dtype)
     if with_mask:
         keep = torch.ones(B, S, device=DEV)
         keep[:, S // 2:] = 0  # mask out the tail keys
         keep[0] = 1
-        mask = ((1.0 - keep[:, None, None, :]) * -10000.0).to(torch.bfloat16)
+        mask = ((1.0 - keep[:, None, None, :]) * -10000.0).to(dtype)
     else:
         mask = torch.Tensor().to(DEV)
     o, lse = e.flash_attn_qkv_fwd(qkv, mask, nh, 1.0 / math.sqrt(hd), 0.0,
