@@ -1,0 +1,147 @@
+"""Latency-oriented inference engine — the serving path grown from the
+reference's ``predict.py`` (single-text, batch-1 inference across
+checkpoints; reference predict.py:104-136).
+
+MI355X design: batch-1 BERT forward is LAUNCH-bound (hundreds of small
+kernels, each ~2-20 us); the whole forward is captured once per input shape
+into a hipGraph and replayed into pinned static buffers afterwards — one
+launch per request. Shapes are bucketed to the next power-of-two sequence
+length so a handful of graphs serve arbitrary inputs. Falls back to plain
+eager forward on CPU or when capture is unavailable.
+"""
+
+from __future__ import annotations
+
+from typing import Dict, List, Optional, Tuple
+
+import torch
+
+
+def _bucket(n: int, lo: int = 32, hi: int = 512) -> int:
+    b = lo
+    while b < n and b < hi:
+        b *= 2
+    return b
+
+
+class _Graphed:
+    """One captured forward for a fixed (batch, seq) shape."""
+
+    def __init__(self, model, batch: int, seq: int, device):
+        self.ids = torch.zeros(batch, seq, dtype=torch.long, device=device)
+        self.mask = torch.zeros(batch, seq, dtype=torch.long, device=device)
+        self.type_ids = torch.zeros(batch, seq, dtype=torch.long, device=device)
+        # warm up on a side stream, then capture
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(2):
+                model(input_ids=self.ids, attention_mask=self.mask,
+                      token_type_ids=self.type_ids)
+        torch.cuda.current_stream().wait_stream(s)
+        self.graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph):
+            out = model(input_ids=self.ids, attention_mask=self.mask,
+                        token_type_ids=self.type_ids)
+            self.logits = out.logits
+
+    def run(self, ids, mask, type_ids):
+        self.ids.copy_(ids, non_blocking=True)
+        self.mask.copy_(mask, non_blocking=True)
+        self.type_ids.copy_(type_ids, non_blocking=True)
+        self.graph.replay()
+        return self.logits
+
+
+class InferenceEngine:
+    """``engine = InferenceEngine(model, tokenizer); engine.predict(texts)``.
+
+    On GPU, each (bucketed) input shape gets a hipGraph-captured forward;
+    on CPU it is a plain no-grad forward.
+    """
+
+    def __init__(self, model, tokenizer=None, device: Optional[str] = None,
+                 max_seq_len: int = 128, use_graph: bool = True):
+        self.device = torch.device(
+            device or ("cuda" if torch.cuda.is_available() else "cpu"))
+        self.model = model.to(self.device).eval()
+        self.tokenizer = tokenizer
+        self.max_seq_len = max_seq_len
+        self.use_graph = use_graph and self.device.type == "cuda"
+        self._graphs: Dict[Tuple[int, int], _Graphed] = {}
+
+    def _encode(self, texts: List[str]):
+        assert self.tokenizer is not None, "predict(texts) needs a tokenizer"
+        ids_l, mask_l, type_l = [], [], []
+        maxlen = 0
+        enc = []
+        for t in texts:
+            ids, mask, type_ids = self.tokenizer.encode(t, self.max_seq_len)
+            n = int(sum(mask))
+            maxlen = max(maxlen, n)
+            enc.append((ids, mask, type_ids))
+        seq = _bucket(maxlen, hi=self.max_seq_len)
+        for ids, mask, type_ids in enc:
+            ids_l.append(ids[:seq])
+            mask_l.append(mask[:seq])
+            type_l.append(type_ids[:seq])
+        return (torch.tensor(ids_l, dtype=torch.long),
+                torch.tensor(mask_l, dtype=torch.long),
+                torch.tensor(type_l, dtype=torch.long))
+
+    @torch.no_grad()
+    def forward_tensors(self, ids, mask, type_ids) -> torch.Tensor:
+        """Logits for already-tokenized fixed-shape inputs."""
+        ids = ids.to(self.device, non_blocking=True)
+        mask = mask.to(self.device, non_blocking=True)
+        type_ids = type_ids.to(self.device, non_blocking=True)
+        if self.use_graph:
+            key = (ids.shape[0], ids.shape[1])
+            g = self._graphs.get(key)
+            if g is None:
+                g = _Graphed(self.model, *key, device=self.device)
+                self._graphs[key] = g
+            return g.run(ids, mask, type_ids)
+        out = self.model(input_ids=ids, attention_mask=mask,
+                         token_type_ids=type_ids)
+        return out.logits
+
+    @torch.no_grad()
+    def predict(self, texts: List[str], id2label: Optional[dict] = None):
+        ids, mask, type_ids = self._encode(texts)
+        logits = self.forward_tensors(ids, mask, type_ids)
+        pred = logits.float().argmax(-1).cpu().tolist()
+        if id2label:
+            return [id2label[int(p)] for p in pred]
+        return pred
+
+    @torch.no_grad()
+    def latency_bench(self, batch: int = 1, seq: Optional[int] = None,
+                      iters: int = 100, warmup: int = 20) -> dict:
+        """Measure request latency for a fixed shape; returns ms stats."""
+        import time
+        seq = seq or self.max_seq_len
+        vocab = getattr(self.model.config, "vocab_size", 21128)
+        g = torch.Generator().manual_seed(0)
+        ids = torch.randint(106, vocab, (batch, seq), generator=g)
+        mask = torch.ones_like(ids)
+        type_ids = torch.zeros_like(ids)
+        lat = []
+        for i in range(warmup + iters):
+            if self.device.type == "cuda":
+                torch.cuda.synchronize()
+            t0 = time.perf_counter()
+            self.forward_tensors(ids, mask, type_ids)
+            if self.device.type == "cuda":
+                torch.cuda.synchronize()
+            dt = (time.perf_counter() - t0) * 1e3
+            if i >= warmup:
+                lat.append(dt)
+        lat.sort()
+        return {
+            "batch": batch, "seq": seq, "iters": iters,
+            "p50_ms": round(lat[len(lat) // 2], 3),
+            "p99_ms": round(lat[int(len(lat) * 0.99) - 1], 3),
+            "mean_ms": round(sum(lat) / len(lat), 3),
+            "graph": self.use_graph,
+        }

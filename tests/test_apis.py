@@ -129,3 +129,21 @@ def test_metrics_jsonl_schema(tmp_path):
     lines = [json.loads(l) for l in open(tmp_path / "met.jsonl")]
     train_lines = [l for l in lines if l.get("phase") == "train"]
     assert train_lines and all("loss" in l and "step" in l for l in train_lines)
+
+
+def test_inference_engine_cpu(tiny_cfg):
+    """Serving path: InferenceEngine eager fallback on CPU (predict +
+    latency_bench smoke; graph capture is covered by the GPU test)."""
+    from pdnlp_amd.engine import InferenceEngine
+    from pdnlp_amd.models import BertForSequenceClassification
+    from pdnlp_amd.data import build_tokenizer
+
+    model = BertForSequenceClassification(tiny_cfg)
+    tok = build_tokenizer(None, vocab_size=tiny_cfg.vocab_size)
+    eng = InferenceEngine(model, tok, device="cpu",
+                          max_seq_len=tiny_cfg.max_position_embeddings)
+    preds = eng.predict(["你好世界", "今天天气不错"])
+    assert len(preds) == 2
+    assert all(0 <= p < tiny_cfg.num_labels for p in preds)
+    stats = eng.latency_bench(batch=1, seq=16, iters=3, warmup=1)
+    assert stats["p50_ms"] > 0 and stats["graph"] is False
