@@ -108,7 +108,10 @@ class HFStyleTrainer:
         eval_loader, _ = self._loader(
             self.eval_dataset, self.hf_args.per_device_eval_batch_size, False)
         loss, acc = self.engine.dev(eval_loader)
-        metrics = {"eval_loss": loss, "eval_accuracy": acc}
+        # engine.dev returns the reference's sum-of-batch-mean losses
+        # (SURVEY.md §2.2); HF Trainer reports the MEAN eval loss
+        metrics = {"eval_loss": loss / max(len(eval_loader), 1),
+                   "eval_accuracy": acc}
         rank0_print(metrics)
         return metrics
 
