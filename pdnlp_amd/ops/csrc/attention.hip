@@ -335,8 +335,8 @@ void fa_bwd_pre_kernel(const T* __restrict__ dout, const T* __restrict__ o,
 template <typename T, typename V8, bool HAS_MASK, bool DROP, bool PRELOAD>
 __global__ __launch_bounds__(NTHREADS)
 void fa_bwd_dq_kernel(const T* __restrict__ dout, const T* __restrict__ qkv,
-                      const float* __restrict__ lse,
-                      const float* __restrict__ dvec,
+                      const T* __restrict__ o, const float* __restrict__ lse,
+                      float* __restrict__ dvec,
                       const T* __restrict__ mask, T* __restrict__ dqkv,
                       const unsigned long long* __restrict__ seed_base,
                       unsigned long long salt, float scale, float p,
@@ -350,6 +350,7 @@ void fa_bwd_dq_kernel(const T* __restrict__ dout, const T* __restrict__ qkv,
   const T* kbase = qbase + H;
   const T* vbase = qbase + 2 * H;
   const T* dobase = dout + b * S * H + h * 64;
+  const T* obase = o + b * S * H + h * 64;
   T* dqbase = dqkv + b * S * ld + h * 64;
 
   __shared__ __attribute__((aligned(16))) char q_lds[64 * 128];
@@ -357,6 +358,7 @@ void fa_bwd_dq_kernel(const T* __restrict__ dout, const T* __restrict__ qkv,
   __shared__ __attribute__((aligned(16))) char k_lds[2][64 * 128];
   __shared__ __attribute__((aligned(16))) char v_lds[2][64 * 128];
   __shared__ __attribute__((aligned(16))) char ds_lds[64 * 128];
+  __shared__ float dv_s[4][16];
 
   const int lane = threadIdx.x & (WAVE - 1);
   const int wid = threadIdx.x >> 6;
@@ -367,6 +369,9 @@ void fa_bwd_dq_kernel(const T* __restrict__ dout, const T* __restrict__ qkv,
 
   stage64<T>(qbase, ld, rb * 64, S, q_lds);
   stage64<T>(dobase, H, rb * 64, S, do_lds);
+  // the dS buffer is free until the KV loop: stage O through it to compute
+  // Dvec = rowsum(dO * O) here (replaces the separate fa_bwd_pre kernel)
+  stage64<T>(obase, H, rb * 64, S, ds_lds);
   stage64<T>(kbase, ld, 0, S, k_lds[0]);
   stage64<T>(vbase, ld, 0, S, v_lds[0]);
   if (PRELOAD && S > 64) {
@@ -376,13 +381,43 @@ void fa_bwd_dq_kernel(const T* __restrict__ dout, const T* __restrict__ qkv,
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
 
+  {
+    // lane l&15 accumulates row wr+(l&15) over its quarter's 8 columns
+    float acc = 0.f;
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      V8 dv = read_n<V8>(do_lds, wr, ks);
+      V8 ov = read_n<V8>(ds_lds, wr, ks);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+        float a, bvv;
+        if constexpr (std::is_same<V8, bf16x8>::value) {
+          a = (float)((__bf16*)&dv)[i];
+          bvv = (float)((__bf16*)&ov)[i];
+        } else {
+          a = (float)((_Float16*)&dv)[i];
+          bvv = (float)((_Float16*)&ov)[i];
+        }
+        acc += a * bvv;
+      }
+    }
+    acc += __shfl_xor(acc, 16, WAVE);
+    acc += __shfl_xor(acc, 32, WAVE);
+    if ((lane >> 4) == 0) dv_s[wid][lane & 15] = acc;
+  }
+  // waits for dv_s AND guards ds_lds reuse by the KV loop
+  __syncthreads();
   float lse_r[4], dvec_r[4];
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
-    const long grow = rb * 64 + wr + (lane >> 4) * 4 + r;
+    const int prow = wr + (lane >> 4) * 4 + r;
+    const long grow = rb * 64 + prow;
     lse_r[r] = lse[(long)bh * S + grow];
-    dvec_r[r] = dvec[(long)bh * S + grow];
+    dvec_r[r] = dv_s[prow >> 4][prow & 15];
   }
+  // one lane per row publishes Dvec for the dK/dV pass (same stream)
+  if ((lane >> 4) == 0)
+    dvec[(long)bh * S + rb * 64 + wr + (lane & 15)] = dv_s[wid][lane & 15];
 
   f32x4 acc_dq[4] = {};
   const int nt = S / 64;
@@ -468,8 +503,9 @@ void fa_bwd_dq_kernel(const T* __restrict__ dout, const T* __restrict__ qkv,
 template <typename T, typename V8, bool HAS_MASK, bool DROP, bool PRELOAD>
 __global__ __launch_bounds__(NTHREADS)
 void fa_bwd_dkv_kernel(const T* __restrict__ dout, const T* __restrict__ qkv,
+                       const T* __restrict__ /*o: consumed by the dQ pass*/,
                        const float* __restrict__ lse,
-                       const float* __restrict__ dvec,
+                       float* __restrict__ dvec,
                        const T* __restrict__ mask, T* __restrict__ dqkv,
                        const unsigned long long* __restrict__ seed_base,
                        unsigned long long salt, float scale, float p,
@@ -685,8 +721,9 @@ static void check_fa_args(const torch::Tensor& qkv, const torch::Tensor& mask,
                        dim3(NTHREADS), 0, stream,                              \
                        (const scalar_t*)dout.data_ptr(),                       \
                        (const scalar_t*)qkv.data_ptr(),                        \
+                       (const scalar_t*)o.data_ptr(),                          \
                        (const float*)lse.data_ptr(),                           \
-                       (const float*)dvec.data_ptr(),                          \
+                       (float*)dvec.data_ptr(),                                \
                        has_mask ? (const scalar_t*)mask.data_ptr() : nullptr,  \
                        (scalar_t*)dqkv.data_ptr(), seed_ptr,                   \
                        (unsigned long long)salt, (float)scale, (float)p,       \
@@ -696,8 +733,9 @@ static void check_fa_args(const torch::Tensor& qkv, const torch::Tensor& mask,
                        dim3(NTHREADS), 0,                                      \
                      stream, (const scalar_t*)dout.data_ptr(),                 \
                      (const scalar_t*)qkv.data_ptr(),                          \
+                     (const scalar_t*)o.data_ptr(),                            \
                      (const float*)lse.data_ptr(),                             \
-                     (const float*)dvec.data_ptr(),                            \
+                     (float*)dvec.data_ptr(),                                  \
                      has_mask ? (const scalar_t*)mask.data_ptr() : nullptr,    \
                      (scalar_t*)dqkv.data_ptr(), seed_ptr,                     \
                      (unsigned long long)salt, (float)scale, (float)p,         \
@@ -756,11 +794,8 @@ torch::Tensor flash_attn_qkv_bwd(torch::Tensor dout, torch::Tensor qkv,
     if constexpr (!std::is_same<scalar_t, float>::value) {
       using V8 = std::conditional_t<
           std::is_same<scalar_t, __hip_bfloat16>::value, bf16x8, f16x8>;
-      hipLaunchKernelGGL((fa_bwd_pre_kernel<scalar_t>),
-                         dim3((total + 3) / 4), dim3(NTHREADS), 0, stream,
-                         (const scalar_t*)dout.data_ptr(),
-                         (const scalar_t*)o.data_ptr(),
-                         (float*)dvec.data_ptr(), (int)nh, (int)S, total);
+      // Dvec = rowsum(dO*O) is computed inside the dQ pass (it already
+      // stages dO; O goes through the dS buffer) and consumed by dK/dV
       dim3 grid(S / 64, B * nh);
       if (has_mask && drop) {
         FA_BWD(fa_bwd_dq_kernel, true, true);
