@@ -309,26 +309,6 @@ void fa_fwd_kernel(const T* __restrict__ qkv, const T* __restrict__ mask,
 }
 
 // ---------------------------------------------------------------------------
-// backward: Dvec = rowsum(dO * O) — one wave per (b, s, h) row, D=64 lanes
-// ---------------------------------------------------------------------------
-
-template <typename T>
-__global__ __launch_bounds__(NTHREADS)
-void fa_bwd_pre_kernel(const T* __restrict__ dout, const T* __restrict__ o,
-                       float* __restrict__ dvec, int nh, int S, long total) {
-  const long idx = (long)blockIdx.x * 4 + (threadIdx.x >> 6);
-  if (idx >= total) return;
-  const int lane = threadIdx.x & (WAVE - 1);
-  const long H = (long)nh * 64;
-  const long bh = idx / S, srow = idx % S;
-  const long b = bh / nh, h = bh % nh;
-  const long off = (b * S + srow) * H + h * 64 + lane;
-  const float v = to_f32<T>(dout[off]) * to_f32<T>(o[off]);
-  const float sum = wave_sum(v);
-  if (lane == 0) dvec[idx] = sum;
-}
-
-// ---------------------------------------------------------------------------
 // backward dQ: grid over 64-row blocks; recompute P from lse, stream K/V
 // ---------------------------------------------------------------------------
 
@@ -788,7 +768,6 @@ torch::Tensor flash_attn_qkv_bwd(torch::Tensor dout, torch::Tensor qkv,
   auto stream = at::hip::getCurrentHIPStream();
   const auto* seed_ptr =
       drop ? (const unsigned long long*)seed_buf.data_ptr() : nullptr;
-  const long total = B * nh * S;
 
   DISPATCH_FLOAT_TYPES(qkv.scalar_type(), "flash_attn_qkv_bwd", [&] {
     if constexpr (!std::is_same<scalar_t, float>::value) {
