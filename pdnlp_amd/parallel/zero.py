@@ -83,6 +83,11 @@ class ZeroRedundancyOptimizer:
             if g.params:
                 self._flatten_group(g)
                 self.groups.append(g)
+        # torch-optimizer-compatible param_groups (GradScaler.unscale_ walks
+        # it; LR schedulers mutate group["lr"] — step() reads it back)
+        self.param_groups = [
+            {"params": g.params, "lr": lr, "betas": betas, "eps": eps,
+             "weight_decay": g.weight_decay} for g in self.groups]
 
     def _flatten_group(self, g: _FlatGroup):
         dev = g.params[0].device
@@ -111,13 +116,13 @@ class ZeroRedundancyOptimizer:
     @torch.no_grad()
     def step(self, closure=None, grad_scale_inv: float = 1.0):
         self.step_count += 1
-        for g in self.groups:
+        for g, pg in zip(self.groups, self.param_groups):
             lo = self.rank * g.shard_size
             shard_grad = self._reduce_scatter(g)
             multi_tensor_adamw(
                 [g.param_flat[lo:lo + g.shard_size]], [shard_grad],
                 [g.m_shard], [g.v_shard], [g.master_shard],
-                self.lr, self.betas[0], self.betas[1], self.eps,
+                pg["lr"], self.betas[0], self.betas[1], self.eps,
                 g.weight_decay, self.step_count, grad_scale_inv)
             self._all_gather(g)
 

@@ -332,7 +332,7 @@ def _attn_ref(qkv, mask, nh, p=0.0):
     return torch.matmul(p_, v).transpose(1, 2).reshape(B, S, H)
 
 
-@pytest.mark.parametrize("S,nh", [(128, 12), (512, 4), (64, 2)])
+@pytest.mark.parametrize("S,nh", [(128, 12), (512, 4), (64, 2), (192, 2)])
 @pytest.mark.parametrize("with_mask", [True, False])
 @pytest.mark.parametrize("dtype", [torch.bfloat16, torch.float16])
 def test_flash_attn_fwd(S, nh, with_mask, dtype):
@@ -356,7 +356,7 @@ def test_flash_attn_fwd(S, nh, with_mask, dtype):
     assert torch.isfinite(lse).all()
 
 
-@pytest.mark.parametrize("S,nh", [(128, 12), (512, 4)])
+@pytest.mark.parametrize("S,nh", [(128, 12), (512, 4), (192, 2)])
 def test_flash_attn_bwd(S, nh):
     e = ext()
     torch.manual_seed(1)

@@ -108,3 +108,30 @@ def _zero_ckpt(rank, world, tmpdir):
 
 def test_zero_sharded_checkpoint(tmp_path):
     run_distributed(_zero_ckpt, world=2, args=(str(tmp_path),))
+
+
+def test_zero_with_grad_scaler_world1():
+    """AMP + ZeRO: GradScaler.unscale_ walks optimizer.param_groups —
+    regression test for the missing attribute."""
+    import torch
+    from pdnlp_amd.amp import GradScaler
+    from pdnlp_amd.parallel.zero import ZeroRedundancyOptimizer
+
+    torch.manual_seed(0)
+    model = torch.nn.Sequential(torch.nn.Linear(16, 16), torch.nn.Linear(16, 4))
+    opt = ZeroRedundancyOptimizer(model, lr=1e-2)
+    scaler = GradScaler(init_scale=8.0)
+    x = torch.randn(4, 16)
+    loss = model(x).square().mean()
+    scaler.scale(loss).backward()
+    scaler.unscale_(opt)
+    assert not scaler._found_inf
+    scaler.step(opt)
+    scaler.update()
+    opt.zero_grad()
+    # LR scheduler contract: mutating param_groups["lr"] must take effect
+    for pg in opt.param_groups:
+        pg["lr"] = 5e-3
+    loss = model(x).square().mean()
+    loss.backward()
+    opt.step()
