@@ -303,6 +303,139 @@ void gemm_nt_pipe_kernel(const T* __restrict__ A, const T* __restrict__ W,
   }
 }
 
+
+// Register-staged BK=128 variant (8 waves, ONE 64-KB LDS buffer): the next
+// K-tile is loaded global->registers while the current tile computes from
+// LDS (T14 issue-early/write-late), then written LDS after a barrier. Twice
+// the MFMA work per barrier pair vs the BK=64 loop — the structure Tensile's
+// winning MT128x128x128 configs use on these small-K shapes.
+template <typename T, typename V8, bool HAS_BIAS, int ACT, bool SAVE_PRE>
+__global__ __launch_bounds__(512)
+void gemm_nt_rs_kernel(const T* __restrict__ A, const T* __restrict__ W,
+                       const T* __restrict__ bias, T* __restrict__ C,
+                       T* __restrict__ pre, long M, long N, long K,
+                       int tiles_n, int nwg) {
+  constexpr int BM = 128, BN = 128, BKR = 128;
+  constexpr int WM = 2, WN = 4;
+  constexpr int TM = BM / WM, TN = BN / WN;   // 64 x 32
+  constexpr int RM = TM / 16, RN = TN / 16;   // 4 x 2
+
+  int wg = blockIdx.x;
+  {
+    const int nxcd = 8;
+    const int q = nwg / nxcd, r = nwg % nxcd;
+    const int xcd = wg % nxcd, idx = wg / nxcd;
+    wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  }
+  const long tile_m = wg / tiles_n, tile_n = wg % tiles_n;
+  const long m0 = tile_m * BM, n0 = tile_n * BN;
+
+  __shared__ __attribute__((aligned(16))) char lds_a[BM * 256];
+  __shared__ __attribute__((aligned(16))) char lds_b[BN * 256];
+
+  const int wid = threadIdx.x >> 6;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wr = (wid / WN) * TM, wc = (wid % WN) * TN;
+
+  // staging geometry: wave wid owns rows [wid*16, wid*16+16) of each tile;
+  // pass p covers 4 rows (16 lanes x 16 B per row = one 256-B row quarter...
+  // actually 64 lanes = 4 rows x 16 lanes x 16 B); 4 passes per tile.
+  const int st_row = (lane >> 4);          // 0..3 within a pass
+  const int st_byte = (lane & 15) * 16;    // 16-B piece of the 256-B row
+  typedef unsigned int uint4v __attribute__((ext_vector_type(4)));
+  uint4v rega[4], regb[4];
+
+  auto load_tile_regs = [&](long k0) {
+#pragma unroll
+    for (int pss = 0; pss < 4; ++pss) {
+      const int r = wid * 16 + pss * 4 + st_row;
+      long gra = m0 + r < M ? m0 + r : M - 1;
+      long grb = n0 + r < N ? n0 + r : N - 1;
+      rega[pss] = *reinterpret_cast<const uint4v*>(
+          (const char*)(A + gra * K + k0) + st_byte);
+      regb[pss] = *reinterpret_cast<const uint4v*>(
+          (const char*)(W + grb * K + k0) + st_byte);
+    }
+  };
+  auto write_tile_lds = [&]() {
+#pragma unroll
+    for (int pss = 0; pss < 4; ++pss) {
+      const int r = wid * 16 + pss * 4 + st_row;
+      const int byte = st_byte ^ ((r & 7) << 4);
+      *reinterpret_cast<uint4v*>(lds_a + r * 256 + byte) = rega[pss];
+      *reinterpret_cast<uint4v*>(lds_b + r * 256 + byte) = regb[pss];
+    }
+  };
+  auto read_frag256 = [&](const char* lds, int frag_row0, int ks) {
+    const int row = frag_row0 + (lane & 15);
+    const int colbyte = (ks * 64 + (lane >> 4) * 16) ^ ((row & 7) << 4);
+    return *reinterpret_cast<const V8*>(lds + row * 256 + colbyte);
+  };
+
+  f32x4 acc[RM][RN] = {};
+
+  load_tile_regs(0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  write_tile_lds();
+  __syncthreads();
+
+  const int ntiles = (int)(K / BKR);
+  for (int t = 0; t < ntiles; ++t) {
+    if (t + 1 < ntiles) load_tile_regs((long)(t + 1) * BKR);
+#pragma unroll
+    for (int ks = 0; ks < 4; ++ks) {
+      V8 a_frag[RM], b_frag[RN];
+#pragma unroll
+      for (int i = 0; i < RM; ++i)
+        a_frag[i] = read_frag256(lds_a, wr + i * 16, ks);
+#pragma unroll
+      for (int j = 0; j < RN; ++j)
+        b_frag[j] = read_frag256(lds_b, wc + j * 16, ks);
+#pragma unroll
+      for (int i = 0; i < RM; ++i) {
+#pragma unroll
+        for (int j = 0; j < RN; ++j) {
+          if constexpr (std::is_same<V8, bf16x8>::value) {
+            acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
+          } else {
+            acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_f16(
+                a_frag[i], b_frag[j], acc[i][j], 0, 0, 0);
+          }
+        }
+      }
+    }
+    if (t + 1 < ntiles) {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __syncthreads();            // everyone done READING the buffer
+      write_tile_lds();
+      __syncthreads();            // buffer refilled
+    }
+  }
+
+  const int crow_off = (lane >> 4) * 4;
+  const int ccol = lane & 15;
+#pragma unroll
+  for (int i = 0; i < RM; ++i) {
+#pragma unroll
+    for (int j = 0; j < RN; ++j) {
+      const long n = n0 + wc + j * 16 + ccol;
+      if (n >= N) continue;
+      const float bv = HAS_BIAS ? to_f32<T>(bias[n]) : 0.f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const long m = m0 + wr + i * 16 + crow_off + r;
+        if (m >= M) continue;
+        float v = acc[i][j][r] + bv;
+        if (SAVE_PRE) pre[m * N + n] = from_f32<T>(v);
+        if (ACT == ACT_GELU) v = gelu_f2(v);
+        if (ACT == ACT_TANH) v = tanhf(v);
+        C[m * N + n] = from_f32<T>(v);
+      }
+    }
+  }
+}
+
 struct TileChoice { int bm, bn; };
 
 // pick the tile so the grid fills 256 CUs (>= ~2 WGs per CU preferred),
@@ -343,6 +476,12 @@ void launch_gemm(const torch::Tensor& A, const torch::Tensor& W,
   // 8 waves default (swept +8-15% over 4): PDNLP_GEMM_W4 reverts
   const bool w8 = std::getenv("PDNLP_GEMM_W4") == nullptr;
   const bool pipe = std::getenv("PDNLP_GEMM_PIPE") != nullptr;
+  const bool rs = std::getenv("PDNLP_GEMM_RS") != nullptr && K % 128 == 0;
+#define LAUNCH_RS(HB, ACTV, SP)                                                \
+  hipLaunchKernelGGL((gemm_nt_rs_kernel<T, V8, HB, ACTV, SP>), dim3(nwg),      \
+                     dim3(512), 0, stream, (const T*)A.data_ptr(),             \
+                     (const T*)W.data_ptr(), bptr, (T*)C.data_ptr(), pptr,     \
+                     M, N, K, tiles_n, nwg)
 #define LAUNCH_P(HB, ACTV, SP, BMV, BNV)                                       \
   do {                                                                         \
     constexpr int shmem = 3 * (BMV + BNV) * 128;                               \
@@ -371,6 +510,7 @@ void launch_gemm(const torch::Tensor& A, const torch::Tensor& W,
       else LAUNCH_T(HB, ACTV, SP, 64, 128, 4);                                 \
     } else if (tc.bm == 128 && tc.bn == 64) LAUNCH_T(HB, ACTV, SP, 128, 64, 4);\
     else if (tc.bm == 64 && tc.bn == 64) LAUNCH_T(HB, ACTV, SP, 64, 64, 4);    \
+    else if (rs) LAUNCH_RS(HB, ACTV, SP);                                      \
     else if (pipe) LAUNCH_P(HB, ACTV, SP, 128, 128);                           \
     else if (w8) LAUNCH_T(HB, ACTV, SP, 128, 128, 8);                          \
     else LAUNCH_T(HB, ACTV, SP, 128, 128, 4);                                  \
