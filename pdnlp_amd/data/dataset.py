@@ -64,11 +64,20 @@ class SyntheticClsDataset(Dataset):
     """Pre-tokenized synthetic classification samples of the reference shape.
 
     Emits dict batches directly (no tokenizer needed): random token ids in
-    [CLS] ... [SEP] form, full attention mask, zero token-type ids, random
-    label. Deterministic per (seed, index)."""
+    [CLS] ... [SEP] form, full attention mask, zero token-type ids.
+    Deterministic per (seed, index).
+
+    ``learnable=True`` (default) makes the label a function of the token
+    content (the bucketed second token id): a correctly-wired model LEARNS
+    it, so epoch runs on synthetic data show real loss curves and dev
+    accuracy — an end-to-end numerics check of every kernel in the stack
+    (random labels would hide a broken gradient anywhere). ``learnable=
+    False`` gives i.i.d. random labels (pure-throughput shape)."""
 
     def __init__(self, n: int, seq_len: int = 128, vocab_size: int = 21128,
-                 num_labels: int = 6, seed: int = 123, var_len: bool = False):
+                 num_labels: int = 6, seed: int = 123, var_len: bool = False,
+                 learnable: bool = True):
+        self.learnable = learnable
         self.n = n
         self.seq_len = seq_len
         self.vocab_size = vocab_size
@@ -91,7 +100,13 @@ class SyntheticClsDataset(Dataset):
         mask[:L] = 1
         ids = ids * mask  # pad with 0 past L
         ids[0] = 101
-        label = int(torch.randint(0, self.num_labels, (1,), generator=g))
+        if self.learnable:
+            # label encoded in the token content (not position-trivial):
+            # bucket of the second token id
+            label = int(ids[1]) * self.num_labels // self.vocab_size
+            label = min(label, self.num_labels - 1)
+        else:
+            label = int(torch.randint(0, self.num_labels, (1,), generator=g))
         return {
             "input_ids": ids.long(),
             "attention_mask": mask,
