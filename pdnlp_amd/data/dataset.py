@@ -101,10 +101,14 @@ class SyntheticClsDataset(Dataset):
         ids = ids * mask  # pad with 0 past L
         ids[0] = 101
         if self.learnable:
-            # label encoded in the token content (not position-trivial):
-            # bucket of the second token id
-            label = int(ids[1]) * self.num_labels // self.vocab_size
-            label = min(label, self.num_labels - 1)
+            # label encoded in the token content: the second token is drawn
+            # from a SMALL repeated set (4 tokens per class) so its embedding
+            # actually trains within an epoch — a bucket of the full vocab
+            # would show each token <1x per epoch and never learn
+            tok = 106 + int(torch.randint(0, self.num_labels * 4, (1,),
+                                          generator=g))
+            ids[1] = tok
+            label = (tok - 106) % self.num_labels
         else:
             label = int(torch.randint(0, self.num_labels, (1,), generator=g))
         return {
