@@ -186,22 +186,29 @@ def test_inference_engine_graph_gpu():
 
 
 def test_learnable_synthetic_convergence_gpu():
-    """End-to-end numerics: 200 fused-path training steps on a LEARNABLE
-    synthetic task must push train accuracy far above chance — a gradient
-    bug in any kernel (GEMM/attention/LN/epilogue/AdamW) stalls this."""
+    """End-to-end numerics: training on a LEARNABLE synthetic task must push
+    train accuracy far above chance — a gradient bug in any kernel
+    (GEMM/attention/LN/epilogue/AdamW) stalls this. Uses a 2-layer model at
+    FULL BERT-base width (so every HIP kernel runs at its real shapes) —
+    12 layers from random init need LR warmup to move in 200 steps, which
+    would test the schedule, not the kernels (2-layer CPU fp32 reference
+    reaches 1.0 accuracy by step ~90)."""
     from torch.utils.data import DataLoader
+    from pdnlp_amd.config import BertConfig
+    from pdnlp_amd.models import BertForSequenceClassification
     from pdnlp_amd.data import SyntheticClsDataset
     from pdnlp_amd.data.collate import Collate
-    from pdnlp_amd.models import build_model
     from pdnlp_amd.ops.adamw import build_optimizer
     from pdnlp_amd.utils import set_seed
     set_seed(123)
-    ds = SyntheticClsDataset(3200, seq_len=128, learnable=True)
+    cfg = BertConfig.bert_base_chinese()
+    cfg.num_hidden_layers = 2
+    ds = SyntheticClsDataset(6400, seq_len=128, learnable=True)
     loader = DataLoader(ds, batch_size=32, shuffle=True,
                         collate_fn=Collate(None, 128))
-    model = build_model("bert-base").to(torch.bfloat16).to(DEV)
+    model = BertForSequenceClassification(cfg).to(torch.bfloat16).to(DEV)
     model.train()
-    opt = build_optimizer(model, lr=1e-4)
+    opt = build_optimizer(model, lr=3e-4)
     correct = total = 0
     for i, batch in enumerate(loader):
         if i >= 200:
@@ -219,4 +226,4 @@ def test_learnable_synthetic_convergence_gpu():
             correct += (out.logits.argmax(-1) == labels).sum().item()
             total += labels.numel()
     acc = correct / max(total, 1)
-    assert acc > 0.55, f"model failed to learn the synthetic task: acc={acc}"
+    assert acc > 0.8, f"model failed to learn the synthetic task: acc={acc}"
