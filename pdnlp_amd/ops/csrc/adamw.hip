@@ -131,12 +131,21 @@ __global__ void unscale_kernel(UnscaleMeta meta, float* found_inf,
   const long local_block = blockIdx.x - meta.block_prefix[t];
   const long n = meta.numel[t];
   T* g = (T*)meta.g[t];
-  const long base = local_block * BLOCK * ILP + threadIdx.x;
   bool bad = false;
+  const long i0 = (local_block * BLOCK + threadIdx.x) * (long)ILP;
+  if ((((unsigned long)g) & (sizeof(v4_t<T>) - 1)) == 0 && i0 + ILP <= n) {
+    v4_t<T> gv = *reinterpret_cast<const v4_t<T>*>(g + i0);
 #pragma unroll
-  for (int j = 0; j < ILP; ++j) {
-    const long i = base + (long)j * BLOCK;
-    if (i < n) {
+    for (int j = 0; j < ILP; ++j) {
+      const float x = to_f32<T>(reinterpret_cast<const T*>(&gv)[j]) * inv_scale;
+      if (!isfinite(x)) bad = true;
+      reinterpret_cast<T*>(&gv)[j] = from_f32<T>(x);
+    }
+    *reinterpret_cast<v4_t<T>*>(g + i0) = gv;
+  } else {
+    for (int j = 0; j < ILP; ++j) {
+      const long i = i0 + j;
+      if (i >= n) break;
       const float x = to_f32<T>(g[i]) * inv_scale;
       if (!isfinite(x)) bad = true;
       g[i] = from_f32<T>(x);
