@@ -10,7 +10,7 @@ memory-optimization study surface.
 from __future__ import annotations
 
 import contextlib
-from typing import Optional
+
 
 import torch
 import torch.distributed as dist

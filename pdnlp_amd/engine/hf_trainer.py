@@ -11,7 +11,7 @@ key like the reference's HF-Trainer variant
 from __future__ import annotations
 
 import os
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Callable, Optional
 
 import torch

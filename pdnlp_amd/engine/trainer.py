@@ -21,7 +21,7 @@ Deliberate deviations (SURVEY.md §2.2 notes):
 from __future__ import annotations
 
 import time
-from typing import Optional
+
 
 import numpy as np
 import torch
@@ -238,7 +238,6 @@ def classification_report_text(trues, preds, label_names=None) -> str:
                                      zero_division=0)
     except ImportError:
         # minimal fallback: per-class P/R/F1
-        import collections
         lines = ["label\tprec\trecall\tf1\tsupport"]
         classes = sorted(set(list(trues) + list(preds)))
         for c in classes:

@@ -17,7 +17,7 @@ the fused ops directly: embedding+LN (K1), fused-QKV GEMM (K2), attention
 
 from __future__ import annotations
 
-import math
+
 from dataclasses import dataclass
 from typing import Optional
 

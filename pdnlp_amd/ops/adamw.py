@@ -10,8 +10,8 @@ decay per group. Torch fallback uses ``torch._foreach_*``.
 
 from __future__ import annotations
 
-import math
-from typing import Iterable, List, Optional
+
+from typing import List, Optional
 
 import torch
 
