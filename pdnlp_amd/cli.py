@@ -89,6 +89,10 @@ def run_classification(args: Args, do_test: bool = True):
     train_loader, dev_loader, train_sampler = build_dataloaders(
         args, args.world_size, args.rank)
     model, optimizer, scaler, trainer = build_training(args)
+    if getattr(args, "resume", "") and os.path.isfile(args.resume):
+        trainer.load_state(args.resume)
+        rank0_print(f"[pdnlp] resumed from {args.resume} "
+                    f"at step {trainer.global_step}")
     minutes = trainer.train(train_loader, dev_loader, train_sampler)
 
     if do_test:

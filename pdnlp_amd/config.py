@@ -85,6 +85,7 @@ class Args:
     # paths
     model_path: str = "./model_hub/chinese-bert-wwm-ext"
     ckpt_path: str = "./output/model.pt"
+    resume: str = ""     # full-state checkpoint to resume training from
     data_path: str = "./data/train.json"
     output_dir: str = "./output"
 

@@ -50,6 +50,7 @@ class Trainer:
                                      rank=self.rank)
         self.best_acc = -1.0  # first dev eval always checkpoints
         self.global_step = 0
+        self._resume_skip = 0  # batches to fast-forward after load_state
 
     # ------------------------------------------------------------------
     def on_step(self, batch):
@@ -130,6 +131,28 @@ class Trainer:
         else:
             self.optimizer.zero_grad(set_to_none=True)
 
+    # ---- resume (a capability the reference lacks — SURVEY.md §5.4) ----
+    def save_state(self, path: str) -> None:
+        """Full training state: model + optimizer + step counters (+ scale)."""
+        extra = {"global_step": self.global_step, "best_acc": self.best_acc}
+        if self.scaler is not None:
+            extra["scaler_scale"] = self.scaler.get_scale()
+        save_checkpoint(self.model, path, optimizer=self.optimizer,
+                        extra=extra, rank=self.rank)
+        if self.world > 1:
+            dist.barrier()  # nobody resumes from a half-written file
+
+    def load_state(self, path: str) -> None:
+        from ..utils.checkpoint import load_checkpoint
+        extra = load_checkpoint(self.model, path,
+                                map_location=self.device,
+                                optimizer=self.optimizer)
+        self.global_step = int(extra.get("global_step", 0))
+        self.best_acc = float(extra.get("best_acc", -1.0))
+        if self.scaler is not None and "scaler_scale" in extra:
+            self.scaler._scale = float(extra["scaler_scale"])
+        self._resume_skip = self.global_step
+
     # ------------------------------------------------------------------
     def train(self, train_loader, dev_loader=None, train_sampler=None):
         args = self.args
@@ -143,6 +166,10 @@ class Trainer:
             if train_sampler is not None and hasattr(train_sampler, "set_epoch"):
                 train_sampler.set_epoch(epoch)
             for step, batch in enumerate(train_loader, start=1):
+                if self._resume_skip > 0:
+                    # fast-forward the data order to the restored step
+                    self._resume_skip -= 1
+                    continue
                 micro += 1
                 accum_boundary = (micro % max(args.grad_accum_steps, 1) == 0)
                 with TraceRange("forward"):
