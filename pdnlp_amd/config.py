@@ -109,7 +109,8 @@ class Args:
     grad_accum_steps: int = 1
     optimizer: str = "adamw"           # "adamw" | "sgd" (fabric alt-path)
     sgd_momentum: float = 0.9
-    lr_scheduler: str = "none"         # "none" | "cosine"
+    lr_scheduler: str = "none"         # "none" | "cosine" | "warmup_linear"
+    warmup_ratio: float = 0.1          # for warmup_linear
 
     # precision
     amp: bool = False

@@ -256,6 +256,29 @@ class Trainer:
         return total_loss, acc, report
 
 
+class _LambdaLR:
+    """Minimal LambdaLR that also works for optimizers that are not
+    torch.optim.Optimizer subclasses (ZeroRedundancyOptimizer)."""
+
+    def __init__(self, optimizer, fn):
+        self.optimizer = optimizer
+        self.fn = fn
+        self._step = 0
+        self._base = [g["lr"] for g in optimizer.param_groups]
+
+    def step(self):
+        self._step += 1
+        for g, b in zip(self.optimizer.param_groups, self._base):
+            g["lr"] = b * self.fn(self._step)
+
+    def state_dict(self):
+        return {"step": self._step, "base": self._base}
+
+    def load_state_dict(self, sd):
+        self._step = sd["step"]
+        self._base = sd["base"]
+
+
 def classification_report_text(trues, preds, label_names=None) -> str:
     try:
         from sklearn.metrics import classification_report
@@ -338,12 +361,29 @@ def build_training(args, model=None, label_key: str = "label"):
             wrapped = model
 
     lr_scheduler = None
-    if args.lr_scheduler == "cosine":
+    if args.lr_scheduler != "none":
         base = optimizer.optimizer if isinstance(optimizer, DistributedOptimizer) \
             else optimizer
-        if isinstance(base, torch.optim.Optimizer):
+        total = max(getattr(args, "total_step", 0), 1) or 1000
+        if args.lr_scheduler == "cosine" and isinstance(base, torch.optim.Optimizer):
             lr_scheduler = torch.optim.lr_scheduler.CosineAnnealingLR(
-                base, T_max=max(args.total_step, 1) or 1000)
+                base, T_max=total)
+        elif args.lr_scheduler == "warmup_linear":
+            # linear warmup over warmup_ratio of the run, then linear decay —
+            # what a from-scratch deep transformer needs to move at all
+            # (pretrained fine-tuning, the reference's setting, does not).
+            # total_step is only known once train() sees the loader, so the
+            # lambda reads it lazily.
+            def lam(step):
+                tot = max(getattr(args, "total_step", 0), 1) or 1000
+                warm = max(int(tot * getattr(args, "warmup_ratio", 0.1)), 1)
+                if step < warm:
+                    return (step + 1) / warm
+                return max(0.0, (tot - step) / max(tot - warm, 1))
+
+            if isinstance(base, torch.optim.Optimizer) or hasattr(
+                    base, "param_groups"):
+                lr_scheduler = _LambdaLR(base, lam)
 
     trainer = Trainer(args, wrapped, optimizer, device, scaler=scaler,
                       lr_scheduler=lr_scheduler, label_key=label_key)
