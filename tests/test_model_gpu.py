@@ -227,3 +227,49 @@ def test_learnable_synthetic_convergence_gpu():
             total += labels.numel()
     acc = correct / max(total, 1)
     assert acc > 0.8, f"model failed to learn the synthetic task: acc={acc}"
+
+
+def test_trainer_engine_gpu_with_resume(tmp_path):
+    """Engine layer on GPU: build_training('single'), a short train run,
+    full-state save, resume in a fresh engine, continue — all on the fused
+    kernel path."""
+    from torch.utils.data import DataLoader
+    from pdnlp_amd.config import Args, BertConfig
+    from pdnlp_amd.data import SyntheticClsDataset
+    from pdnlp_amd.data.collate import Collate
+    from pdnlp_amd.engine.trainer import Trainer, build_training
+    from pdnlp_amd.models import BertForSequenceClassification
+    from pdnlp_amd.utils import set_seed
+
+    set_seed(123)
+    cfg = BertConfig.bert_base_chinese()
+    cfg.num_hidden_layers = 2
+    args = Args()
+    args.epochs = 1
+    args.do_dev = False
+    args.log_every = 4
+    args.amp = True
+    args.amp_dtype = "bf16"
+    args.ckpt_path = str(tmp_path / "m.pt")
+
+    def loader():
+        ds = SyntheticClsDataset(256, seq_len=128)
+        return DataLoader(ds, batch_size=16, shuffle=False,
+                          collate_fn=Collate(None, 128))
+
+    model = BertForSequenceClassification(cfg)
+    wrapped, opt, scaler, trainer = build_training(args, model=model)
+    trainer.train(loader())
+    assert trainer.global_step == 16
+    st = str(tmp_path / "state.pt")
+    trainer.save_state(st)
+
+    model2 = BertForSequenceClassification(cfg)
+    w2, o2, s2, tr2 = build_training(args, model=model2)
+    tr2.load_state(st)
+    assert tr2.global_step == 16
+    tr2.train(loader())          # resume skips all 16 -> no new steps
+    assert tr2.global_step == 16
+    for (n, a), (_, b) in zip(wrapped.state_dict().items(),
+                              w2.state_dict().items()):
+        torch.testing.assert_close(a, b, rtol=0, atol=0)
