@@ -26,7 +26,6 @@ torch::Tensor bias_gelu_fwd(torch::Tensor x, torch::Tensor b);
 std::vector<torch::Tensor> bias_gelu_bwd(torch::Tensor dy, torch::Tensor x,
                                          torch::Tensor b);
 torch::Tensor gelu_bwd(torch::Tensor dy, torch::Tensor pre);
-std::vector<torch::Tensor> gelu_bwd_dbias(torch::Tensor dy, torch::Tensor pre);
 torch::Tensor col_sum(torch::Tensor x);
 torch::Tensor gemm_tn(torch::Tensor A, torch::Tensor B);
 torch::Tensor tanh_bwd(torch::Tensor dy, torch::Tensor pre);
@@ -73,7 +72,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bias_gelu_fwd", &bias_gelu_fwd);
   m.def("bias_gelu_bwd", &bias_gelu_bwd);
   m.def("gelu_bwd", &gelu_bwd);
-  m.def("gelu_bwd_dbias", &gelu_bwd_dbias);
   m.def("col_sum", &col_sum);
   m.def("gemm_tn", &gemm_tn);
   m.def("tanh_bwd", &tanh_bwd);

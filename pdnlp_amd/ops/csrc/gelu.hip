@@ -114,40 +114,6 @@ __global__ void tanh_bwd_kernel(const T* __restrict__ dy,
 }
 
 
-// dgelu with the FFN-up bias gradient fused: thread owns 4 fixed columns,
-// loops a row chunk, register-accumulates db, one atomicAdd per column per
-// chunk — replaces gelu_bwd + a separate torch column-reduce (SURVEY.md K7
-// backward).
-template <typename T>
-__global__ __launch_bounds__(256)
-void gelu_bwd_dbias_kernel(const T* __restrict__ dy, const T* __restrict__ pre,
-                           T* __restrict__ dx, float* __restrict__ db32,
-                           long R, int N, long rows_per_chunk) {
-  const int c0 = (blockIdx.x * blockDim.x + threadIdx.x) * 4;
-  if (c0 >= N) return;
-  const long r0 = blockIdx.y * rows_per_chunk;
-  const long r1 = min(r0 + rows_per_chunk, R);
-  float db[4] = {};
-#pragma unroll 4
-  for (long r = r0; r < r1; ++r) {
-    const short4 dv = *reinterpret_cast<const short4*>(dy + r * N + c0);
-    const short4 pv = *reinterpret_cast<const short4*>(pre + r * N + c0);
-    short4 ov;
-    const T* pd = reinterpret_cast<const T*>(&dv);
-    const T* pp = reinterpret_cast<const T*>(&pv);
-    T* po = reinterpret_cast<T*>(&ov);
-#pragma unroll
-    for (int j = 0; j < 4; ++j) {
-      const float g = to_f32<T>(pd[j]) * gelu_grad_f(to_f32<T>(pp[j]));
-      po[j] = from_f32<T>(g);
-      db[j] += g;
-    }
-    *reinterpret_cast<short4*>(dx + r * N + c0) = ov;
-  }
-#pragma unroll
-  for (int j = 0; j < 4; ++j) atomicAdd(db32 + c0 + j, db[j]);
-}
-
 // plain fp32 column sum of a bf16/fp16 [R, N] matrix (projection-bias
 // gradients for the fused-QKV GEMM: replaces torch's reduce_kernel)
 template <typename T>
@@ -236,30 +202,6 @@ torch::Tensor tanh_bwd(torch::Tensor dy, torch::Tensor pre) {
                        (scalar_t*)dx.data_ptr(), total);
   });
   return dx;
-}
-
-std::vector<torch::Tensor> gelu_bwd_dbias(torch::Tensor dy, torch::Tensor pre) {
-  TORCH_CHECK(dy.dim() == 2 || dy.dim() == 3, "gelu_bwd_dbias: 2D/3D input");
-  const int N = dy.size(-1);
-  const long R = dy.numel() / N;
-  TORCH_CHECK(N % 4 == 0 && dy.scalar_type() != torch::kFloat,
-              "gelu_bwd_dbias: bf16/fp16, N % 4 == 0");
-  auto dx = torch::empty_like(dy);
-  auto db32 = torch::zeros({(long)N}, dy.options().dtype(torch::kFloat32));
-  auto stream = at::hip::getCurrentHIPStream();
-  const long rows_per_chunk = 64;
-  const long chunks = (R + rows_per_chunk - 1) / rows_per_chunk;
-  dim3 grid((N / 4 + 255) / 256, chunks);
-  DISPATCH_FLOAT_TYPES(dy.scalar_type(), "gelu_bwd_dbias", [&] {
-    if constexpr (!std::is_same<scalar_t, float>::value) {
-      hipLaunchKernelGGL((gelu_bwd_dbias_kernel<scalar_t>), grid, dim3(256), 0,
-                         stream, (const scalar_t*)dy.data_ptr(),
-                         (const scalar_t*)pre.data_ptr(),
-                         (scalar_t*)dx.data_ptr(), db32.data_ptr<float>(), R,
-                         N, rows_per_chunk);
-    }
-  });
-  return {dx, db32.to(dy.scalar_type())};
 }
 
 torch::Tensor col_sum(torch::Tensor x) {

@@ -152,3 +152,29 @@ def test_ddp_grads_match_reference_world4():
     """8-GPU-shaped sanity at world=4 (gloo): bucket launch order and
     views stay correct as the bucket count interacts with more ranks."""
     run_distributed(_ddp_grads_match_reference, world=4)
+
+
+def _ddp_grad_compression(rank, world):
+    import torch
+    from pdnlp_amd.parallel.ddp import DistributedDataParallel
+    torch.manual_seed(123)
+    model = torch.nn.Sequential(torch.nn.Linear(32, 32), torch.nn.Linear(32, 8))
+    ddp = DistributedDataParallel(model, bucket_cap_mb=0.001,
+                                  grad_compression="bf16")
+    torch.manual_seed(500 + rank)
+    x = torch.randn(4, 32)
+    ddp(x).square().mean().backward()
+    ddp.finalize_backward()
+    # reference: bf16-compressed average of per-rank grads
+    import torch.distributed as dist
+    g = model[0].weight.grad.clone()
+    gather = [torch.zeros_like(g) for _ in range(world)]
+    dist.all_gather(gather, g)
+    for other in gather:
+        assert torch.allclose(g, other), "ranks disagree after allreduce"
+
+
+def test_ddp_grad_compression_bf16():
+    """Horovod-style wire compression on the DDP reducer (SURVEY C6 folded
+    into C1): fp32 grads, bf16 on the wire, ranks converge identically."""
+    run_distributed(_ddp_grad_compression, world=2)

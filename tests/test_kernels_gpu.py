@@ -429,18 +429,10 @@ def test_flash_attn_dropout_bwd_matches_masked_ref():
     assert abs(drop_rate - p) < 0.05
 
 
-def test_gelu_bwd_dbias_and_col_sum():
+def test_col_sum():
     e = ext()
     torch.manual_seed(5)
-    R, N = 2048, 3072
-    dy = torch.randn(R, N, device=DEV, dtype=torch.bfloat16)
-    pre = torch.randn(R, N, device=DEV, dtype=torch.bfloat16)
-    dx, db = e.gelu_bwd_dbias(dy, pre)
-    ref_pre = pre.float().detach().requires_grad_()
-    F.gelu(ref_pre).backward(dy.float())
-    torch.testing.assert_close(dx.float(), ref_pre.grad, rtol=2e-2, atol=2e-2)
-    torch.testing.assert_close(db.float(), ref_pre.grad.sum(0),
-                               rtol=2e-2, atol=2e-1)
+    dy = torch.randn(2048, 3072, device=DEV, dtype=torch.bfloat16)
     cs = e.col_sum(dy)
     torch.testing.assert_close(cs.float(), dy.float().sum(0),
                                rtol=2e-2, atol=2e-1)
