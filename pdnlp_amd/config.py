@@ -110,6 +110,7 @@ class Args:
     optimizer: str = "adamw"           # "adamw" | "sgd" (fabric alt-path)
     sgd_momentum: float = 0.9
     lr_scheduler: str = "none"         # "none" | "cosine" | "warmup_linear"
+    hip_graph: bool = False            # capture fwd+bwd into a hipGraph (1 GPU)
     warmup_ratio: float = 0.1          # for warmup_linear
 
     # precision

@@ -51,6 +51,9 @@ class Trainer:
         self.best_acc = -1.0  # first dev eval always checkpoints
         self.global_step = 0
         self._resume_skip = 0  # batches to fast-forward after load_state
+        self._graph = None
+        self._graph_warm = 0
+        self._static = None
 
     # ------------------------------------------------------------------
     def on_step(self, batch):
@@ -131,6 +134,71 @@ class Trainer:
         else:
             self.optimizer.zero_grad(set_to_none=True)
 
+    # ---- hipGraph-captured training step (single GPU, SURVEY.md §5.1+) ----
+    def _maybe_capture_graph(self, batch):
+        """Capture fwd+bwd into a hipGraph after the first eager steps.
+
+        Enabled by ``args.hip_graph`` on a single GPU without grad
+        accumulation: inputs are copied into static device buffers and the
+        graph replayed — kernel-launch-free steps. Grads are captured with
+        set_to_none semantics so every replay OVERWRITES the same blocks
+        (no zero_grad, no accumulate-adds; see bench.py). The optimizer
+        stays eager; dropout reseeds per replay via the device seed."""
+        if self._graph is not None:
+            return True
+        if not (getattr(self.args, "hip_graph", False)
+                and torch.cuda.is_available() and self.world == 1
+                and getattr(self.args, "grad_accum_steps", 1) <= 1
+                and self.scaler is None):
+            return False
+        if self._graph_warm < 3:   # eager warmup steps before capture
+            self._graph_warm += 1
+            return False
+        keys = ("input_ids", "attention_mask", "token_type_ids",
+                self.label_key)
+        self._static = {k: batch[k].to(self.device).clone() for k in keys}
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(2):
+                out = self.model(
+                    input_ids=self._static["input_ids"],
+                    attention_mask=self._static["attention_mask"],
+                    token_type_ids=self._static["token_type_ids"],
+                    labels=self._static[self.label_key])
+                out.loss.backward()
+                self.optimizer.step()
+        torch.cuda.current_stream().wait_stream(s)
+        self.optimizer.zero_grad(set_to_none=True)
+        self._graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self._graph):
+            out = self.model(
+                input_ids=self._static["input_ids"],
+                attention_mask=self._static["attention_mask"],
+                token_type_ids=self._static["token_type_ids"],
+                labels=self._static[self.label_key])
+            out.loss.backward()
+            self._static_loss = out.loss
+            self._static_logits = out.logits
+        rank0_print("[pdnlp] hipGraph captured: replaying fwd+bwd")
+        return True
+
+    def _graph_step(self, batch):
+        from ..ops import reseed_dropout
+        for k, t in self._static.items():
+            if batch[k].shape != t.shape:
+                return None  # tail batch with a different shape: run eager
+        for k, t in self._static.items():
+            t.copy_(batch[k].to(self.device, non_blocking=True),
+                    non_blocking=True)
+        reseed_dropout()
+        self._graph.replay()
+        self.optimizer.step()   # grads overwritten in-place by the replay
+        if self.lr_scheduler is not None:
+            self.lr_scheduler.step()
+        return (self._static_loss, self._static_logits,
+                self._static[self.label_key])
+
     # ---- resume (a capability the reference lacks — SURVEY.md §5.4) ----
     def save_state(self, path: str) -> None:
         """Full training state: model + optimizer + step counters (+ scale)."""
@@ -172,6 +240,16 @@ class Trainer:
                     continue
                 micro += 1
                 accum_boundary = (micro % max(args.grad_accum_steps, 1) == 0)
+                if self._maybe_capture_graph(batch):
+                    res = self._graph_step(batch)
+                    if res is not None:
+                        loss, logits, labels = res
+                        stepped = True
+                        self.global_step += 1
+                        timer.step(labels.shape[0] * self.world)
+                        self._after_step(epoch, total_step, loss, timer,
+                                         dev_loader, stepped)
+                        continue
                 with TraceRange("forward"):
                     loss, logits, labels = self.on_step(batch)
                 if args.barrier_per_step and self.world > 1:
@@ -179,35 +257,8 @@ class Trainer:
                 stepped = self._backward_and_step(loss, accum_boundary)
                 self.global_step += 1
                 timer.step(labels.shape[0] * self.world)
-                if self.global_step % args.log_every == 0:
-                    # the .item() device sync happens only on logged steps
-                    if self.global_step % args.loss_reduce_every == 0:
-                        printed_loss = self.loss_reduce(loss).item()
-                    else:
-                        printed_loss = loss.item()
-                    rank0_print(
-                        f"【train】 epoch：{epoch}/{args.epochs} "
-                        f"step：{self.global_step}/{total_step} "
-                        f"loss：{printed_loss:.6f}")
-                    self.metrics.write(
-                        phase="train", epoch=epoch, step=self.global_step,
-                        loss=printed_loss,
-                        lr=self.optimizer.param_groups[0]["lr"]
-                        if hasattr(self.optimizer, "param_groups") else args.learning_rate,
-                        samples_per_sec=timer.samples_per_sec(),
-                        stepped=stepped)
-                if (dev_loader is not None and args.do_dev
-                        and self.global_step % args.eval_step == 0):
-                    dev_loss, acc = self.dev(dev_loader)
-                    rank0_print(f"【dev】 loss：{dev_loss:.6f} accuracy：{acc:.4f}")
-                    self.metrics.write(phase="dev", step=self.global_step,
-                                       loss=dev_loss, accuracy=acc)
-                    if acc > self.best_acc:
-                        self.best_acc = acc
-                        save_checkpoint(self.model, args.ckpt_path, rank=self.rank)
-                        rank0_print(f"【best】 accuracy：{acc:.4f} → saved "
-                                    f"{args.ckpt_path}")
-                    self.model.train()
+                self._after_step(epoch, total_step, loss, timer, dev_loader,
+                                 stepped)
         if not (args.do_dev and dev_loader is not None):
             save_checkpoint(self.model, args.ckpt_path, rank=self.rank)
         mins = (time.time() - t_start) / 60.0
@@ -215,6 +266,39 @@ class Trainer:
         self.metrics.write(phase="train_end", minutes=mins,
                            samples_per_sec=timer.samples_per_sec(sync=True))
         return mins
+
+    def _after_step(self, epoch, total_step, loss, timer, dev_loader,
+                    stepped):
+        args = self.args
+        if self.global_step % args.log_every == 0:
+            # the .item() device sync happens only on logged steps
+            if self.global_step % args.loss_reduce_every == 0:
+                printed_loss = self.loss_reduce(loss).item()
+            else:
+                printed_loss = loss.item()
+            rank0_print(
+                f"【train】 epoch：{epoch}/{args.epochs} "
+                f"step：{self.global_step}/{total_step} "
+                f"loss：{printed_loss:.6f}")
+            self.metrics.write(
+                phase="train", epoch=epoch, step=self.global_step,
+                loss=printed_loss,
+                lr=self.optimizer.param_groups[0]["lr"]
+                if hasattr(self.optimizer, "param_groups") else args.learning_rate,
+                samples_per_sec=timer.samples_per_sec(),
+                stepped=stepped)
+        if (dev_loader is not None and args.do_dev
+                and self.global_step % args.eval_step == 0):
+            dev_loss, acc = self.dev(dev_loader)
+            rank0_print(f"【dev】 loss：{dev_loss:.6f} accuracy：{acc:.4f}")
+            self.metrics.write(phase="dev", step=self.global_step,
+                               loss=dev_loss, accuracy=acc)
+            if acc > self.best_acc:
+                self.best_acc = acc
+                save_checkpoint(self.model, args.ckpt_path, rank=self.rank)
+                rank0_print(f"【best】 accuracy：{acc:.4f} → saved "
+                            f"{args.ckpt_path}")
+            self.model.train()
 
     # ------------------------------------------------------------------
     @torch.no_grad()
