@@ -168,6 +168,9 @@ class Trainer:
                     labels=self._static[self.label_key])
                 out.loss.backward()
                 self.optimizer.step()
+        del out  # a live autograd graph from before the capture pins
+        #          default/side-stream AccumulateGrad nodes and SEGFAULTS
+        #          hipGraph instantiation (torch input_buffer stream check)
         torch.cuda.current_stream().wait_stream(s)
         self.optimizer.zero_grad(set_to_none=True)
         self._graph = torch.cuda.CUDAGraph()
@@ -240,6 +243,9 @@ class Trainer:
                     continue
                 micro += 1
                 accum_boundary = (micro % max(args.grad_accum_steps, 1) == 0)
+                # drop the previous step's autograd references BEFORE any
+                # capture attempt (see note in _maybe_capture_graph)
+                loss = logits = None
                 if self._maybe_capture_graph(batch):
                     res = self._graph_step(batch)
                     if res is not None:
