@@ -67,9 +67,21 @@ def main():
         model = build_model(args.model, model_path=args.model_path)
         if c is not None:
             load_checkpoint(model, c)
-        model = model.to(device)
-        pred = predict(model, text, tokenizer, args.max_seq_len, device)
-        print(f"{c or '<random-init>'} → predicted: {LABELS[pred]}")
+        if ns.engine == "graphed":
+            from pdnlp_amd.engine import InferenceEngine
+            if torch.cuda.is_available():
+                model = model.to(torch.bfloat16)
+            eng = InferenceEngine(model, tokenizer, device=str(device),
+                                  max_seq_len=args.max_seq_len)
+            pred = eng.predict([text])[0]
+            stats = eng.latency_bench(batch=1, seq=args.max_seq_len,
+                                      iters=30, warmup=10)
+            print(f"{c or '<random-init>'} → predicted: {LABELS[pred]} "
+                  f"(p50 {stats['p50_ms']} ms, graphed={stats['graph']})")
+        else:
+            model = model.to(device)
+            pred = predict(model, text, tokenizer, args.max_seq_len, device)
+            print(f"{c or '<random-init>'} → predicted: {LABELS[pred]}")
 
 
 if __name__ == "__main__":
