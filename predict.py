@@ -41,6 +41,10 @@ def main():
     p.add_argument("--label", type=int, default=3,
                    help="pick a random dataset sample with this label")
     p.add_argument("--ckpt", action="append", default=None)
+    p.add_argument("--engine", choices=["eager", "graphed"],
+                   default="eager",
+                   help="graphed = hipGraph-captured serving forward "
+                        "(pdnlp_amd.engine.InferenceEngine)")
     ns, rest = p.parse_known_args()
     args = Args().apply_cli(rest)
     set_seed(args.seed)
