@@ -320,3 +320,34 @@ def test_trainer_hip_graph_mode_gpu(tmp_path):
             correct += (out.logits.argmax(-1).cpu() == b["label"]).sum().item()
             total += b["label"].numel()
     assert correct / total > 0.8, f"graphed training failed: {correct/total}"
+
+
+def test_activation_checkpointing_gpu():
+    """Activation checkpointing (+CPU offload) on the fused path: memory
+    drops, grads still flow (the DeepSpeed-config capability)."""
+    from pdnlp_amd.models import build_model
+    from pdnlp_amd.utils import set_seed
+
+    def peak_mem(ckpt, offload=False):
+        set_seed(3)
+        torch.cuda.empty_cache()
+        torch.cuda.reset_peak_memory_stats()
+        model = build_model("bert-base").to(torch.bfloat16).to(DEV)
+        if ckpt:
+            model.gradient_checkpointing_enable(cpu_offload=offload)
+        model.train()
+        g = torch.Generator().manual_seed(0)
+        ids = torch.randint(106, 21128, (32, 128), generator=g).to(DEV)
+        out = model(input_ids=ids, attention_mask=torch.ones_like(ids),
+                    labels=torch.randint(0, 6, (32,), generator=g).to(DEV))
+        out.loss.backward()
+        torch.cuda.synchronize()
+        n_grads = sum(p.grad is not None for p in model.parameters())
+        assert n_grads > 100, "grads missing under checkpointing"
+        return torch.cuda.max_memory_allocated()
+
+    base = peak_mem(False)
+    ck = peak_mem(True)
+    off = peak_mem(True, offload=True)
+    assert ck < base, (base, ck)
+    assert off <= ck * 1.05, (ck, off)
