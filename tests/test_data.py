@@ -57,3 +57,28 @@ def test_distributed_sampler_sharding():
     s0.set_epoch(1)
     e1 = list(s0)
     assert e0 != e1
+
+
+def test_distributed_sampler_exact_torch_parity():
+    """Our DistributedSampler must produce EXACTLY torch's indices (same
+    seed/epoch semantics) so runs are comparable with the reference."""
+    import torch
+    from torch.utils.data import TensorDataset
+    from torch.utils.data.distributed import DistributedSampler as TorchDS
+    from pdnlp_amd.data.sampler import DistributedSampler as OurDS
+
+    for n in (100, 96, 7):
+        ds = TensorDataset(torch.arange(n))
+        for world in (1, 2, 3):
+            for rank in range(world):
+                for shuffle in (True, False):
+                    for epoch in (0, 1, 5):
+                        a = TorchDS(ds, num_replicas=world, rank=rank,
+                                    shuffle=shuffle, seed=0)
+                        b = OurDS(ds, num_replicas=world, rank=rank,
+                                  shuffle=shuffle, seed=0)
+                        a.set_epoch(epoch)
+                        b.set_epoch(epoch)
+                        assert list(a) == list(b), (n, world, rank, shuffle,
+                                                    epoch)
+                        assert len(a) == len(b)
