@@ -83,5 +83,6 @@ from .functional import (  # noqa: F401,E402
     bias_dropout_residual_layernorm,
     dropout,
     reseed_dropout,
+    dw_stream_join,
 )
 from .adamw import FusedAdamW, multi_tensor_adamw  # noqa: F401,E402

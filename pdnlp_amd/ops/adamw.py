@@ -73,6 +73,8 @@ class FusedAdamW(torch.optim.Optimizer):
     @torch.no_grad()
     def step(self, closure=None, grad_scale_inv: float = 1.0,
              found_inf: Optional[torch.Tensor] = None):
+        from .functional import dw_stream_join
+        dw_stream_join()   # no-op unless PDNLP_DW_STREAM=1
         loss = None
         if closure is not None:
             with torch.enable_grad():

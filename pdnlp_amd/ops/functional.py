@@ -86,6 +86,49 @@ def embedding_layernorm(input_ids, token_type_ids, position_ids,
 # backward dGEMMs (plain, no fusion) go through rocBLAS via torch.matmul.
 # --------------------------------------------------------------------------
 
+class _DwStream:
+    """Optional side stream for the dW GEMMs (PDNLP_DW_STREAM=1).
+
+    dW = dY^T·X does not feed the backward chain (only dX does), so it can
+    overlap the next layer's backward. Lifetime/order handling:
+    - the side stream waits the producing stream before each dW;
+    - dY/X are record_stream'd so the allocator cannot recycle them under
+      the in-flight GEMM;
+    - consumers (optimizer step, DDP bucket reduction) call ``join()``
+      which makes the current stream wait the side stream ONCE.
+    Default OFF: single-GPU eager measured within noise of the serialized
+    order on the flagship shape (the dGEMM grid already fills the chip), so
+    the extra ordering surface is not enabled blindly; kept as a measured
+    option for shapes where backward has launch gaps.
+    """
+
+    def __init__(self):
+        self.stream = None
+
+    def enabled(self) -> bool:
+        import os
+        return (os.environ.get("PDNLP_DW_STREAM", "0") == "1"
+                and torch.cuda.is_available())
+
+    def get(self):
+        if self.stream is None:
+            self.stream = torch.cuda.Stream()
+        return self.stream
+
+    def join(self):
+        if self.stream is not None:
+            torch.cuda.current_stream().wait_stream(self.stream)
+
+
+dw_stream = _DwStream()
+
+
+def dw_stream_join() -> None:
+    """Make the current stream wait all side-stream dW GEMMs (call before
+    consuming gradients when PDNLP_DW_STREAM=1)."""
+    dw_stream.join()
+
+
 class _LinearHipFn(torch.autograd.Function):
     """y = x @ w^T + b with the forward GEMM on the hand-written MFMA kernel
     (optionally fused activation), backward dGEMMs on rocBLAS."""
@@ -110,7 +153,16 @@ class _LinearHipFn(torch.autograd.Function):
         elif ctx.act == "tanh":
             dy2 = ext().tanh_bwd(dy2, pre)
         dx = dy2 @ w                      # rocBLAS NN
-        dw = dy2.t() @ x2                 # rocBLAS TN
+        if dw_stream.enabled():
+            s = dw_stream.get()
+            s.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(s):
+                dw = dy2.t() @ x2         # off the backward critical path
+            dy2.record_stream(s)
+            x2.record_stream(s)
+            dw.record_stream(torch.cuda.current_stream())
+        else:
+            dw = dy2.t() @ x2             # rocBLAS TN
         if ctx.has_bias and db is None:
             if dy2.shape[-1] % 4 == 0 and dy2.shape[0] >= 256 \
                     and dy2.dtype != torch.float32:

@@ -144,14 +144,19 @@ class DistributedDataParallel(nn.Module):
 
     def _launch(self, b: _Bucket):
         scale = 1.0 / self.world_size if self.average else 1.0
+        from ..ops.functional import dw_stream
         if self._comm_stream is not None:
             cur = torch.cuda.current_stream()
             self._comm_stream.wait_stream(cur)
+            if dw_stream.stream is not None:
+                # side-stream dW GEMMs must land before the bucket reduces
+                self._comm_stream.wait_stream(dw_stream.stream)
             with torch.cuda.stream(self._comm_stream):
                 self._reduce_bucket(b, scale)
             b.event = torch.cuda.Event()
             b.event.record(self._comm_stream)
         else:
+            dw_stream.join()
             self._reduce_bucket(b, scale)
 
     def _reduce_bucket(self, b: _Bucket, scale: float):

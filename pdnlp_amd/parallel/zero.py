@@ -115,6 +115,8 @@ class ZeroRedundancyOptimizer:
     # ------------------------------------------------------------------
     @torch.no_grad()
     def step(self, closure=None, grad_scale_inv: float = 1.0):
+        from ..ops.functional import dw_stream_join
+        dw_stream_join()   # no-op unless PDNLP_DW_STREAM=1
         self.step_count += 1
         for g, pg in zip(self.groups, self.param_groups):
             lo = self.rank * g.shard_size
