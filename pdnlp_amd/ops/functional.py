@@ -96,10 +96,11 @@ class _DwStream:
       the in-flight GEMM;
     - consumers (optimizer step, DDP bucket reduction) call ``join()``
       which makes the current stream wait the side stream ONCE.
-    Default OFF: single-GPU eager measured within noise of the serialized
-    order on the flagship shape (the dGEMM grid already fills the chip), so
-    the extra ordering surface is not enabled blindly; kept as a measured
-    option for shapes where backward has launch gaps.
+    Default OFF — and measured SLOWER on the flagship shape (4109 -> 3373
+    samples/s): the dGEMM grids already fill all 256 CUs, so the "overlap"
+    just time-slices CUs while the per-layer stream switches and waits add
+    real cost. Kept as an env-gated experiment with the measurement
+    recorded (DESIGN.md negative results).
     """
 
     def __init__(self):
