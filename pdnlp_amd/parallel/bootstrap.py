@@ -42,6 +42,10 @@ def init_distributed(backend: Optional[str] = None,
         return local
 
     backend = backend or _default_backend()
+    # a crashed rank must ABORT peers' collectives instead of hanging them
+    # (the reference simply hangs); the watchdog honors the
+    # same timeout passed to init_process_group below
+    os.environ.setdefault("TORCH_NCCL_ASYNC_ERROR_HANDLING", "1")
     # bind the device BEFORE creating the process group so RCCL communicators
     # are built against the right HIP device (one process per GPU over xGMI)
     if torch.cuda.is_available():
