@@ -178,3 +178,8 @@ def test_ddp_grad_compression_bf16():
     """Horovod-style wire compression on the DDP reducer (SURVEY C6 folded
     into C1): fp32 grads, bf16 on the wire, ranks converge identically."""
     run_distributed(_ddp_grad_compression, world=2)
+
+
+def test_ddp_grads_match_reference_world3():
+    """Odd world size: uneven sharding + bucket math."""
+    run_distributed(_ddp_grads_match_reference, world=3)
