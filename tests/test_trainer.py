@@ -158,3 +158,22 @@ def test_resume_matches_uninterrupted_run(tmp_path, tiny_cfg):
                                 model_c.named_parameters()):
         torch.testing.assert_close(pa, pc, rtol=0, atol=0,
                                    msg=lambda m: f"{n}: {m}")
+
+
+def test_multi_epoch_reshuffles(tmp_path, tiny_cfg):
+    """epochs=2: the sampler reshuffles per epoch and steps double."""
+    from pdnlp_amd.data.sampler import DistributedSampler
+    args, loader, trainer = _setup(tmp_path, tiny_cfg)
+    args.epochs = 2
+    args.do_dev = False
+    ds = loader.dataset
+    sampler = DistributedSampler(ds, num_replicas=1, rank=0, shuffle=True)
+    sampler.set_epoch(1)
+    order1 = list(sampler)
+    sampler.set_epoch(2)
+    order2 = list(sampler)
+    assert order1 != order2, "epochs must reshuffle"
+    loader2 = DataLoader(ds, batch_size=8, sampler=sampler,
+                         collate_fn=loader.collate_fn)
+    trainer.train(loader2, train_sampler=sampler)
+    assert trainer.global_step == 2 * len(loader2)
