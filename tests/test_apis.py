@@ -147,3 +147,25 @@ def test_inference_engine_cpu(tiny_cfg):
     assert all(0 <= p < tiny_cfg.num_labels for p in preds)
     stats = eng.latency_bench(batch=1, seq=16, iters=3, warmup=1)
     assert stats["p50_ms"] > 0 and stats["graph"] is False
+
+
+def test_inference_engine_batching_and_buckets(tiny_cfg):
+    """Shape bucketing: variable-length inputs land in pow-2 buckets; batch
+    predictions match single predictions."""
+    from pdnlp_amd.engine import InferenceEngine
+    from pdnlp_amd.engine.infer import _bucket
+    from pdnlp_amd.models import BertForSequenceClassification
+    from pdnlp_amd.data import build_tokenizer
+
+    assert _bucket(1) == 32 and _bucket(33) == 64 and _bucket(600, hi=512) == 512
+    model = BertForSequenceClassification(tiny_cfg).eval()
+    tok = build_tokenizer(None, vocab_size=tiny_cfg.vocab_size)
+    eng = InferenceEngine(model, tok, device="cpu",
+                          max_seq_len=tiny_cfg.max_position_embeddings)
+    texts = ["你好", "今天天气不错啊朋友", "第三个句子"]
+    batch_preds = eng.predict(texts)
+    single_preds = [eng.predict([t])[0] for t in texts]
+    # single predictions may use a smaller bucket; recompute batch at the
+    # same per-text buckets for comparison is overkill — lengths here all
+    # bucket to 32, so they must agree exactly
+    assert batch_preds == single_preds
