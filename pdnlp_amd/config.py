@@ -111,6 +111,8 @@ class Args:
     sgd_momentum: float = 0.9
     lr_scheduler: str = "none"         # "none" | "cosine" | "warmup_linear"
     hip_graph: bool = False            # capture fwd+bwd into a hipGraph (1 GPU)
+    torch_profile_steps: int = 0       # >0: profile that many steps (after 3
+                                       # warmup) to output_dir/trace.json
     warmup_ratio: float = 0.1          # for warmup_linear
 
     # precision

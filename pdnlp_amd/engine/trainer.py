@@ -233,6 +233,28 @@ class Trainer:
         timer = StepTimer(self.device)
         t_start = time.time()
         micro = 0
+        prof = None
+        if getattr(args, "torch_profile_steps", 0) > 0 and self.rank == 0:
+            # first-class tracing the reference lacks:
+            # kernel-level chrome trace for a few steps, written to
+            # output_dir/trace.json
+            import os
+            from torch.profiler import (ProfilerActivity, profile, schedule,
+                                        tensorboard_trace_handler)  # noqa: F401
+            out = os.path.join(getattr(args, "output_dir", "."),
+                               "trace.json")
+
+            def _export(p):
+                p.export_chrome_trace(out)
+                rank0_print(f"[pdnlp] torch.profiler trace -> {out}")
+
+            prof = profile(
+                activities=[ProfilerActivity.CPU, ProfilerActivity.CUDA],
+                schedule=schedule(wait=1, warmup=2,
+                                  active=args.torch_profile_steps),
+                on_trace_ready=_export)
+            prof.__enter__()
+        self._prof = prof
         for epoch in range(1, args.epochs + 1):
             if train_sampler is not None and hasattr(train_sampler, "set_epoch"):
                 train_sampler.set_epoch(epoch)
@@ -265,6 +287,9 @@ class Trainer:
                 timer.step(labels.shape[0] * self.world)
                 self._after_step(epoch, total_step, loss, timer, dev_loader,
                                  stepped)
+        if prof is not None:
+            prof.__exit__(None, None, None)
+        self._prof = None
         if not (args.do_dev and dev_loader is not None):
             save_checkpoint(self.model, args.ckpt_path, rank=self.rank)
         mins = (time.time() - t_start) / 60.0
@@ -276,6 +301,8 @@ class Trainer:
     def _after_step(self, epoch, total_step, loss, timer, dev_loader,
                     stepped):
         args = self.args
+        if getattr(self, "_prof", None) is not None:
+            self._prof.step()
         if self.global_step % args.log_every == 0:
             # the .item() device sync happens only on logged steps
             if self.global_step % args.loss_reduce_every == 0:

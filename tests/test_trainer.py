@@ -177,3 +177,13 @@ def test_multi_epoch_reshuffles(tmp_path, tiny_cfg):
                          collate_fn=loader.collate_fn)
     trainer.train(loader2, train_sampler=sampler)
     assert trainer.global_step == 2 * len(loader2)
+
+
+def test_torch_profiler_trace(tmp_path, tiny_cfg):
+    """--torch-profile-steps exports a chrome trace (SURVEY 5.1 tracing)."""
+    args, loader, trainer = _setup(tmp_path, tiny_cfg)
+    args.do_dev = False
+    args.torch_profile_steps = 1
+    args.output_dir = str(tmp_path)
+    trainer.train(loader)
+    assert (tmp_path / "trace.json").exists()
