@@ -66,7 +66,8 @@ def main():
     if world > 1:
         from pdnlp_amd.parallel.bootstrap import init_distributed
         init_distributed()
-    device = torch.device(f"cuda:{local_rank}" if use_cuda else "cpu")
+    dev_idx = local_rank % max(torch.cuda.device_count(), 1) if use_cuda else 0
+    device = torch.device(f"cuda:{dev_idx}" if use_cuda else "cpu")
     if use_cuda:
         torch.cuda.set_device(device)
 
