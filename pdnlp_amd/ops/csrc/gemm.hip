@@ -509,7 +509,10 @@ void launch_gemm(const torch::Tensor& A, const torch::Tensor& W,
       else if (w8) LAUNCH_T(HB, ACTV, SP, 64, 128, 8);                         \
       else LAUNCH_T(HB, ACTV, SP, 64, 128, 4);                                 \
     } else if (tc.bm == 128 && tc.bn == 64) LAUNCH_T(HB, ACTV, SP, 128, 64, 4);\
-    else if (tc.bm == 64 && tc.bn == 64) LAUNCH_T(HB, ACTV, SP, 64, 64, 4);    \
+    else if (tc.bm == 64 && tc.bn == 64) {                                     \
+      if (w8) LAUNCH_T(HB, ACTV, SP, 64, 64, 8);                               \
+      else LAUNCH_T(HB, ACTV, SP, 64, 64, 4);                                  \
+    }                                                                          \
     else if (rs) LAUNCH_RS(HB, ACTV, SP);                                      \
     else if (pipe) LAUNCH_P(HB, ACTV, SP, 128, 128);                           \
     else if (w8) LAUNCH_T(HB, ACTV, SP, 128, 128, 8);                          \
