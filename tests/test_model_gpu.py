@@ -419,3 +419,26 @@ def test_bert_large_width_seq512_numerics_vs_fp32():
     diff = (got.logits.float().cpu() - ref.logits).abs().max().item()
     assert diff < 0.15, f"seq512 bf16 drift {diff}"
     assert abs(got.loss.item() - ref.loss.item()) < 0.05
+
+
+def test_non_multiple_of_64_seq_falls_back():
+    """S=100 (not a multiple of 64) must route attention to the batched
+    GEMM + fused-masked-softmax fallback and still match fp32."""
+    from pdnlp_amd.config import BertConfig
+    from pdnlp_amd.models import BertForSequenceClassification
+    from pdnlp_amd.utils import set_seed
+    set_seed(123)
+    cfg = BertConfig.bert_base_chinese()
+    cfg.num_hidden_layers = 2
+    cfg.hidden_dropout_prob = 0.0
+    cfg.attention_probs_dropout_prob = 0.0
+    model = BertForSequenceClassification(cfg).eval()
+    ids, mask, type_ids, labels = _batch(cfg.vocab_size, cfg.num_labels,
+                                         B=2, S=100)
+    with torch.no_grad():
+        ref = model(ids, mask, type_ids, labels)
+    gm = model.to(torch.bfloat16).to(DEV)
+    with torch.no_grad():
+        got = gm(ids.to(DEV), mask.to(DEV), type_ids.to(DEV), labels.to(DEV))
+    diff = (got.logits.float().cpu() - ref.logits).abs().max().item()
+    assert diff < 0.15, diff
